@@ -1,0 +1,62 @@
+"""Run configuration.
+
+Mirrors the reference CLI surface (G2Vec.py:505-518) and adds the
+MI355X-framework knobs (device, dtype, seed, DP) that default to
+reference-equivalent behaviour.
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import Optional
+
+
+@dataclasses.dataclass
+class G2VecConfig:
+    # --- reference-compatible surface (G2Vec.py:505-518) ---
+    expression_file: str = ""
+    clinical_file: str = ""
+    network_file: str = ""
+    result_name: str = "result"
+    len_path: int = 80          # -p / --lenPath
+    num_repetition: int = 10    # -r / --numRepetition
+    hidden: int = 128           # -s / --sizeHiddenlayer
+    epochs: int = 500           # -e / --epoch (honoured here; ref ignores it, G2Vec.py:262)
+    lr: float = 0.005           # -l / --learningRate
+    num_biomarker: int = 50     # -n / --numBiomarker
+
+    # --- framework knobs (absent in reference) ---
+    seed: Optional[int] = 0         # None -> nondeterministic like the reference
+    dtype: str = "bf16"             # compute dtype of the W_ih gather: {"fp32","bf16"}
+    device: str = "auto"            # "auto" | "cpu" | "cuda"
+    pcc_threshold: float = 0.5      # |PCC| cutoff (G2Vec.py:385-390)
+    pcc_mode: str = "auto"          # "edge" (per-edge dot) | "gemm" (MFMA corr GEMM) | "auto"
+    compat_lgroup_bug: bool = False  # reproduce the shipped G2Vec.py:186-194 behaviour
+    early_stop: bool = True
+    batch_size: int = 0             # 0 = full batch (reference semantics, G2Vec.py:264)
+    trainer_path: str = "fast"      # "fast": collapsed rank-1 path (linear-net algebra)
+                                    # "general": full gather/scatter kernel chain (K1-K8)
+    save_paths: str = ""            # cache generated path set (de-facto checkpoint)
+    load_paths: str = ""
+    log_jsonl: str = ""             # structured metrics sink
+    deterministic_grads: bool = False  # bitwise-reproducible dW_ih reduction (no atomics)
+
+    def validate(self) -> None:
+        if self.hidden % 64 != 0 or not (64 <= self.hidden <= 1024):
+            raise ValueError(
+                f"hidden={self.hidden}: MI355X kernels require a multiple of the "
+                f"64-lane wavefront in [64, 1024]")
+        if self.dtype not in ("fp32", "bf16"):
+            raise ValueError(f"dtype must be fp32|bf16, got {self.dtype}")
+        if self.len_path < 1 or self.len_path > 512:
+            raise ValueError("len_path must be in [1, 512] (LDS visited-list budget)")
+        if self.pcc_mode not in ("auto", "edge", "gemm"):
+            raise ValueError(f"bad pcc_mode {self.pcc_mode}")
+        if self.trainer_path not in ("fast", "general"):
+            raise ValueError(f"bad trainer_path {self.trainer_path}")
+
+
+def resolve_device(device: str) -> str:
+    if device != "auto":
+        return device
+    import torch
+    return "cuda" if torch.cuda.is_available() else "cpu"

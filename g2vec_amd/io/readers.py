@@ -1,0 +1,77 @@
+"""Input parsers for the three tab-delimited G2Vec formats.
+
+Format parity (manual.pdf pp.3-4):
+  expression: header `PATIENT\t<sample...>`, rows `gene\t<float...>`
+              (reference parser: G2Vec.py:478-503; transposed to samples x genes)
+  clinical:   header line skipped, rows `sample\tint` (G2Vec.py:436-453)
+  network:    header line skipped, rows `src\tdest` directed (G2Vec.py:455-476)
+
+When the native extension is importable its C++ parser is used for the
+expression matrix (the only input that gets large at synthetic scale);
+the pure-Python fallback is semantics-identical.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Set, Tuple
+
+import numpy as np
+
+
+def load_expression(path: str) -> dict:
+    """Returns {'sample': [S] str array, 'expr': f32 [S, G], 'gene': [G] str array}."""
+    try:
+        from g2vec_amd import _C  # native fast path
+        genes, samples, expr = _C.parse_expression_tsv(path)
+        return {
+            "sample": np.array(samples),
+            "expr": expr.numpy().astype(np.float32, copy=False),
+            "gene": np.array(genes),
+        }
+    except ImportError:
+        pass
+    with open(path) as fin:
+        header = fin.readline().rstrip("\n").split("\t")
+        samples = header[1:]
+        genes: List[str] = []
+        rows: List[np.ndarray] = []
+        for line in fin:
+            parts = line.rstrip("\n").split("\t")
+            if len(parts) < 2:
+                continue
+            genes.append(parts[0])
+            rows.append(np.asarray(parts[1:], dtype=np.float32))
+    expr = np.stack(rows, axis=0).T  # gene-wise -> sample-wise, like G2Vec.py:498
+    return {"sample": np.array(samples), "expr": expr, "gene": np.array(genes)}
+
+
+def load_clinical(path: str) -> Dict[str, int]:
+    """sample -> label (0 good / 1 poor). Header line skipped (G2Vec.py:449)."""
+    result: Dict[str, int] = {}
+    with open(path) as fin:
+        fin.readline()
+        for line in fin:
+            parts = line.rstrip().split("\t")
+            if len(parts) < 2:
+                continue
+            result[parts[0]] = int(parts[1])
+    return result
+
+
+def load_network(path: str) -> dict:
+    """{'edge': [(src, dst)...] directed, in file order; 'gene': set}.
+
+    The reference keeps the network DIRECTED as-read (G2Vec.py:390 writes
+    adjMat[src][dest] only) — we preserve that.
+    """
+    edges: List[Tuple[str, str]] = []
+    geneset: Set[str] = set()
+    with open(path) as fin:
+        fin.readline()  # header
+        for line in fin:
+            parts = line.rstrip().split("\t")
+            if len(parts) < 2:
+                continue
+            edges.append((parts[0], parts[1]))
+            geneset.add(parts[0])
+            geneset.add(parts[1])
+    return {"edge": edges, "gene": geneset}

@@ -1,0 +1,246 @@
+"""Modified-CBOW trainer (reference step 4, G2Vec.py:217-286).
+
+Model: X (multi-hot paths) @ W_ih -> H; H @ W_ho -> logit; sigmoid-CE on the
+prognosis label. No biases, no nonlinearity (G2Vec.py:238-243) — the net is
+LINEAR, which the MI355X fast path exploits:
+
+  forward:   o_p = sum_{g in p} s_g      with s = W_ih @ W_ho   (one GEMV +
+             a scalar segment-sum kernel instead of a [B,G]x[G,h] matmul)
+  backward:  dW_ih = c (outer) W_ho,  dW_ho = W_ih^T c,
+             with c = X^T dO in R^G    (rank-1 gradient; the DP all-reduce
+             message is G+h floats, not G*h)
+
+The kernel-chain "general" path (gather rows -> wave reduce -> scatter-add,
+SURVEY §2.10 K1-K8) computes the identical math without the collapse and is
+the template for non-linear successors; both paths share Adam and early-stop.
+
+Reference-semantics notes:
+  - training is FULL-BATCH, one Adam step per epoch (G2Vec.py:262-264);
+    --batch-size > 0 opts into minibatching (documented divergence)
+  - accuracy is evaluated with the POST-update weights each epoch
+    (sess.run(optimizer) then acc.eval, G2Vec.py:264-267)
+  - early stop: first epoch whose val-ACC is strictly lower than the
+    previous epoch's; the returned W_ih is the PREVIOUS epoch's
+    (keep-last-good semantics, G2Vec.py:276-283)
+  - Adam is TF1 AdamOptimizer (dense moments; lr_t = lr*sqrt(1-b2^t)/(1-b1^t))
+  - init: truncated normal, stddev 1/sqrt(hidden), +-2 sigma (G2Vec.py:234-235)
+"""
+from __future__ import annotations
+
+import dataclasses
+import time
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+from ..config import G2VecConfig
+from ..parallel.dist import DistContext, single
+from ..paths import PathSet, subset
+
+
+@dataclasses.dataclass
+class TrainResult:
+    W_ih: torch.Tensor            # f32 [G, h] on the training device
+    stop_epoch: int               # reference's reported epoch (step-1 on stop)
+    acc_val: float
+    acc_tr: float
+    epochs_run: int
+    acc_val_history: List[float]
+    epoch_times_s: List[float]
+    wall_to_acc_s: Optional[float]   # wall-clock until val-ACC >= 0.88 (None if never)
+
+
+class CbowTrainer:
+    B1, B2, EPS = 0.9, 0.999, 1e-8   # TF1 AdamOptimizer defaults (G2Vec.py:246)
+    ACC_TARGET = 0.88                # BASELINE.md headline threshold
+
+    def __init__(self, cfg: G2VecConfig, n_genes: int,
+                 device: torch.device, ctx: Optional[DistContext] = None,
+                 log=print):
+        cfg.validate()
+        self.cfg = cfg
+        self.G = n_genes
+        self.h = cfg.hidden
+        self.device = device
+        self.ctx = ctx or single(device)
+        self.log = log
+
+    # ------------------------------------------------------------------ setup
+    def _init_weights(self, gen: Optional[torch.Generator]):
+        std = 1.0 / (self.h ** 0.5)
+        W = torch.empty(self.G, self.h, dtype=torch.float32)
+        who = torch.empty(self.h, dtype=torch.float32)
+        if gen is None:
+            torch.nn.init.trunc_normal_(W, std=std, a=-2 * std, b=2 * std)
+            torch.nn.init.trunc_normal_(who, std=std, a=-2 * std, b=2 * std)
+        else:
+            with torch.no_grad():
+                W.copy_(_trunc_normal(W.shape, std, gen))
+                who.copy_(_trunc_normal(who.shape, std, gen))
+        return W.to(self.device), who.to(self.device)
+
+    def _split(self, ps: PathSet):
+        """Seeded shuffle + 80/20 split (G2Vec.py:219-226), then DP shard."""
+        P = ps.n_paths
+        gen = torch.Generator()
+        if self.cfg.seed is not None:
+            gen.manual_seed(int(self.cfg.seed) + 12345)
+        perm = torch.randperm(P, generator=gen).to(self.device)
+        pivot = int(P * 0.8)
+        tr_idx, vl_idx = perm[:pivot], perm[pivot:]
+        self.n_tr_global = int(tr_idx.numel())
+        self.n_vl_global = int(vl_idx.numel())
+        tr_idx = tr_idx[self.ctx.shard_indices(tr_idx.numel(), self.device)]
+        vl_idx = vl_idx[self.ctx.shard_indices(vl_idx.numel(), self.device)]
+        return subset(ps, tr_idx), subset(ps, vl_idx)
+
+    # ------------------------------------------------------------------ train
+    def train(self, ps: PathSet) -> TrainResult:
+        cfg = self.cfg
+        gen = None
+        if cfg.seed is not None:
+            gen = torch.Generator()
+            gen.manual_seed(int(cfg.seed))
+        W, who = self._init_weights(gen)
+        self.ctx.broadcast_(W)      # C4: replicated params (all ranks identical)
+        self.ctx.broadcast_(who)
+        tr, vl = self._split(ps)
+
+        mW = torch.zeros_like(W)
+        vW = torch.zeros_like(W)
+        mO = torch.zeros_like(who)
+        vO = torch.zeros_like(who)
+        W_keep = W.clone()
+
+        use_general = cfg.trainer_path == "general"
+        W16 = W.bfloat16() if (use_general and cfg.dtype == "bf16") else None
+        plan = None
+        if not use_general:
+            plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)
+
+        inv_b = 1.0 / max(self.n_tr_global, 1)
+        # minibatch boundaries over the local shard (full batch by default)
+        P_loc = tr.n_paths
+        bs = cfg.batch_size if cfg.batch_size > 0 else P_loc
+        batches = [(i, min(i + bs, P_loc)) for i in range(0, max(P_loc, 1), bs)]
+
+        before_val, before_tr = -1.0, -1.0
+        stop_epoch = -1
+        acc_hist: List[float] = []
+        epoch_times: List[float] = []
+        wall_to_acc = None
+        t_adam = 0
+        t0_all = time.perf_counter()
+        display_step = 5
+        blk_t0 = time.perf_counter()
+
+        self.log("     Start training the modified CBOW with early stopping")
+        epochs_run = 0
+        for epoch in range(cfg.epochs):
+            ep_t0 = time.perf_counter()
+            for (lo, hi) in batches:
+                b_inv = inv_b if cfg.batch_size == 0 else 1.0 / (
+                    (hi - lo) * self.ctx.world)
+                t_adam += 1
+                if use_general:
+                    self._step_general(W, W16, who, mW, vW, mO, vO, tr, lo, hi,
+                                       b_inv, t_adam)
+                else:
+                    self._step_fast(W, who, mW, vW, mO, vO, tr, plan, lo, hi,
+                                    b_inv, t_adam)
+            # post-update accuracy on both splits (reference order,
+            # G2Vec.py:264-267) — one fused scalar forward per split
+            acc_tr = self._accuracy(W, W16, who, tr, self.n_tr_global)
+            acc_val = self._accuracy(W, W16, who, vl, self.n_vl_global)
+            epochs_run = epoch + 1
+            acc_hist.append(acc_val)
+            epoch_times.append(time.perf_counter() - ep_t0)
+            if wall_to_acc is None and acc_val >= self.ACC_TARGET:
+                wall_to_acc = time.perf_counter() - t0_all
+
+            if epoch % display_step == 0:
+                self.log("    - Epoch: %03d\tACC[val]=%.4f\tACC[tr]=%.4f (%.3f sec)"
+                         % (epoch, acc_val, acc_tr,
+                            time.perf_counter() - blk_t0))
+                blk_t0 = time.perf_counter()
+            if cfg.early_stop and acc_val < before_val:
+                stop_epoch = epoch - 1
+                self.log("    - Epoch(stop): %03d\tACC[val]=%.4f\tACC[tr]=%.4f (%.3f sec)"
+                         % (stop_epoch, before_val, before_tr,
+                            time.perf_counter() - blk_t0))
+                acc_val, acc_tr = before_val, before_tr
+                break
+            before_val, before_tr = acc_val, acc_tr
+            W_keep.copy_(W)         # keep-last-good snapshot (G2Vec.py:283)
+        self.log("    Optimization Finish")
+
+        return TrainResult(W_ih=W_keep, stop_epoch=stop_epoch,
+                           acc_val=acc_val, acc_tr=acc_tr,
+                           epochs_run=epochs_run, acc_val_history=acc_hist,
+                           epoch_times_s=epoch_times, wall_to_acc_s=wall_to_acc)
+
+    # ------------------------------------------------------------------ steps
+    def _slice(self, ps: PathSet, lo: int, hi: int):
+        if lo == 0 and hi == ps.n_paths:
+            return ps.genes, ps.offsets, ps.labels
+        offs = ps.offsets.long()
+        g = ps.genes[offs[lo]:offs[hi]]
+        o = (ps.offsets[lo:hi + 1] - ps.offsets[lo]).contiguous()
+        return g, o, ps.labels[lo:hi]
+
+    def _step_fast(self, W, who, mW, vW, mO, vO, tr, plan, lo, hi, inv_b, t):
+        genes, offsets, labels = self._slice(tr, lo, hi)
+        s = torch.mv(W, who)
+        _loss, _corr, dO = ops.cbow_fwd_scalar(s, genes, offsets, labels,
+                                               inv_b, True)
+        use_plan = plan if (lo == 0 and hi == tr.n_paths) else None
+        c = ops.scatter_dO(genes, offsets, dO, self.G, plan=use_plan)
+        self.ctx.allreduce_(c)                      # C1: the whole dW_ih message
+        grad_who = torch.mv(W.t(), c)               # dW_ho = W_ih^T c (pre-update W)
+        ops.adam_rank1(W, mW, vW, c, who, t, self.cfg.lr, self.B1, self.B2,
+                       self.EPS)
+        ops.adam_dense(who, mO, vO, grad_who, t, self.cfg.lr, self.B1,
+                       self.B2, self.EPS)
+
+    def _step_general(self, W, W16, who, mW, vW, mO, vO, tr, lo, hi, inv_b, t):
+        genes, offsets, labels = self._slice(tr, lo, hi)
+        Wg = W16 if W16 is not None else W
+        _loss, _corr, dO, H = ops.cbow_fwd(Wg, who, genes, offsets, labels,
+                                           inv_b, True)
+        dW = ops.cbow_bwd_rows(who, genes, offsets, dO, self.G)
+        grad_who = torch.mv(H.t(), dO)
+        self.ctx.allreduce_(dW)
+        self.ctx.allreduce_(grad_who)
+        ops.adam_dense(W, mW, vW, dW, t, self.cfg.lr, self.B1, self.B2, self.EPS)
+        ops.adam_dense(who, mO, vO, grad_who, t, self.cfg.lr, self.B1,
+                       self.B2, self.EPS)
+        if W16 is not None:
+            W16.copy_(W)
+
+    def _accuracy(self, W, W16, who, ps: PathSet, n_global: int) -> float:
+        if ps.n_paths == 0:
+            correct = torch.zeros((), dtype=torch.float32, device=self.device)
+        elif self.cfg.trainer_path == "general":
+            Wg = W16 if W16 is not None else W
+            _l, corr, _d, _h = ops.cbow_fwd(Wg, who, ps.genes, ps.offsets,
+                                            ps.labels, 1.0, False)
+            correct = corr.sum()
+        else:
+            s = torch.mv(W, who)
+            _l, corr, _d = ops.cbow_fwd_scalar(s, ps.genes, ps.offsets,
+                                               ps.labels, 1.0, False)
+            correct = corr.sum()
+        self.ctx.allreduce_(correct)                # C3 scalar metric reduce
+        return float(correct.item()) / max(n_global, 1)
+
+
+def _trunc_normal(shape, std: float, gen: torch.Generator) -> torch.Tensor:
+    """Seeded +-2sigma truncated normal via rejection resampling
+    (reference init semantics, tf.truncated_normal G2Vec.py:234-235)."""
+    out = torch.randn(shape, generator=gen) * std
+    bad = out.abs() > 2 * std
+    while bool(bad.any()):
+        out[bad] = torch.randn((int(bad.sum()),), generator=gen) * std
+        bad = out.abs() > 2 * std
+    return out

@@ -1,0 +1,135 @@
+"""Device-op dispatch layer.
+
+CUDA/HIP tensors -> the in-tree native extension `g2vec_amd._C`
+(hand-written gfx950 HIP kernels; FAILS LOUDLY if the extension is not
+built — there is deliberately no silent eager fallback on a GPU box).
+CPU tensors -> the torch/numpy oracles in `cpu_ref`.
+"""
+from __future__ import annotations
+
+from typing import NamedTuple, Optional, Tuple
+
+import torch
+
+from . import cpu_ref
+
+_NATIVE = None
+_NATIVE_ERR: Optional[BaseException] = None
+try:  # pragma: no cover - exercised only when the extension is built
+    from g2vec_amd import _C as _NATIVE  # type: ignore
+except Exception as e:  # noqa: BLE001
+    _NATIVE_ERR = e
+
+
+def native_available() -> bool:
+    return _NATIVE is not None
+
+
+def native():
+    if _NATIVE is None:
+        raise RuntimeError(
+            "g2vec_amd._C HIP extension is not importable on this host — the GPU "
+            "compute path refuses to fall back to eager PyTorch. Build it in-tree "
+            "with `python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). "
+            f"Import error was: {_NATIVE_ERR!r}")
+    return _NATIVE
+
+
+# ------------------------------------------------------------------ walks
+def random_walks(row_ptr, col_idx, weights, sources, num_repetition: int,
+                 len_path: int, seed: int):
+    if row_ptr.is_cuda:
+        return native().random_walks(row_ptr, col_idx, weights, sources,
+                                     num_repetition, len_path, seed)
+    return cpu_ref.random_walks(row_ptr, col_idx, weights, sources,
+                                num_repetition, len_path, seed)
+
+
+# ------------------------------------------------------------------ CBOW fast (scalar) path
+def cbow_fwd_scalar(s, genes, offsets, labels, inv_b: float, want_grad: bool):
+    if s.is_cuda:
+        loss, correct, dO = native().cbow_fwd_scalar(
+            s, genes, offsets, labels, float(inv_b), bool(want_grad))
+        return loss, correct, (dO if want_grad else None)
+    return cpu_ref.cbow_fwd_scalar(s, genes, offsets, labels, inv_b, want_grad)
+
+
+class ScatterPlan(NamedTuple):
+    """Precomputed gene-sorted instance layout for the deterministic
+    c = X^T dO reduction (built once per path set; genes never change
+    across epochs)."""
+    inst_path: torch.Tensor   # i32 [nnz]  path id of each instance, gene-sorted
+    seg_start: torch.Tensor   # i32 [n_seg+1]
+    seg_gene: torch.Tensor    # i32 [n_seg]
+
+
+def build_scatter_plan(genes: torch.Tensor, offsets: torch.Tensor,
+                       n_genes: int) -> ScatterPlan:
+    counts = (offsets[1:] - offsets[:-1]).long()
+    path_of = torch.repeat_interleave(
+        torch.arange(len(counts), device=genes.device), counts)
+    sorted_genes, perm = torch.sort(genes.long(), stable=True)
+    inst_path = path_of[perm].int().contiguous()
+    seg_gene, seg_counts = torch.unique_consecutive(sorted_genes, return_counts=True)
+    seg_start = torch.zeros(len(seg_gene) + 1, dtype=torch.int64, device=genes.device)
+    torch.cumsum(seg_counts, 0, out=seg_start[1:])
+    return ScatterPlan(inst_path, seg_start.int().contiguous(),
+                       seg_gene.int().contiguous())
+
+
+def scatter_dO(genes, offsets, dO, n_genes: int,
+               plan: Optional[ScatterPlan] = None):
+    if dO.is_cuda:
+        if plan is None:
+            plan = build_scatter_plan(genes, offsets, n_genes)
+        return native().scatter_dO_det(plan.inst_path, plan.seg_start,
+                                       plan.seg_gene, dO, int(n_genes))
+    return cpu_ref.scatter_dO(genes, offsets, dO, n_genes)
+
+
+def adam_rank1(W, m, v, c, who, t: int, lr: float, b1: float, b2: float,
+               eps: float) -> None:
+    if W.is_cuda:
+        native().adam_rank1(W, m, v, c, who, int(t), float(lr), float(b1),
+                            float(b2), float(eps))
+        return
+    cpu_ref.adam_rank1(W, m, v, c, who, t, lr, b1, b2, eps)
+
+
+def adam_dense(W, m, v, grad, t: int, lr: float, b1: float, b2: float,
+               eps: float) -> None:
+    if W.is_cuda:
+        native().adam_dense(W, m, v, grad, int(t), float(lr), float(b1),
+                            float(b2), float(eps))
+        return
+    cpu_ref.adam_dense(W, m, v, grad, t, lr, b1, b2, eps)
+
+
+# ------------------------------------------------------------------ CBOW general (kernel-chain) path
+def cbow_fwd(W, who, genes, offsets, labels, inv_b: float, want_grad: bool):
+    if W.is_cuda:
+        loss, correct, dO, H = native().cbow_fwd(
+            W, who, genes, offsets, labels, float(inv_b), bool(want_grad))
+        if not want_grad:
+            return loss, correct, None, None
+        return loss, correct, dO, H
+    return cpu_ref.cbow_fwd(W, who, genes, offsets, labels, inv_b, want_grad)
+
+
+def cbow_bwd_rows(who, genes, offsets, dO, n_genes: int):
+    if dO.is_cuda:
+        return native().cbow_bwd_rows(who, genes, offsets, dO, int(n_genes))
+    return cpu_ref.cbow_bwd_rows(who, genes, offsets, dO, n_genes)
+
+
+# ------------------------------------------------------------------ graph / PCC
+def pcc_edges(zt, edge_idx, n_group: int):
+    if zt.is_cuda:
+        return native().pcc_edges(zt, edge_idx, int(n_group))
+    return cpu_ref.pcc_edges(zt, edge_idx, n_group)
+
+
+def corr_gemm(zt, n_group: int):
+    if zt.is_cuda:
+        return native().corr_gemm(zt, int(n_group))
+    return cpu_ref.corr_gemm(zt, n_group)
