@@ -1,0 +1,354 @@
+// Torch bindings for the g2vec_amd gfx950 kernels + native host utilities.
+// One translation unit: kernels are included below.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstring>
+#include <fstream>
+#include <sstream>
+#include <string>
+#include <vector>
+
+#include "g2vec_kernels.hip"
+
+#define CHECK_DEV(x) TORCH_CHECK((x).is_cuda(), #x " must be on the GPU")
+#define CHECK_CONT(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+#define CHECK_I32(x) TORCH_CHECK((x).scalar_type() == at::kInt, #x " must be int32")
+#define CHECK_F32(x) TORCH_CHECK((x).scalar_type() == at::kFloat, #x " must be float32")
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+inline int grid_for(long long work_items, int per_block) {
+  long long blocks = (work_items + per_block - 1) / per_block;
+  if (blocks > (1 << 20)) blocks = 1 << 20;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+#define LAUNCH_CHECK()                                                       \
+  do {                                                                       \
+    hipError_t err = hipGetLastError();                                      \
+    TORCH_CHECK(err == hipSuccess, "HIP launch failed: ",                    \
+                hipGetErrorString(err));                                     \
+  } while (0)
+
+// ------------------------------------------------------------------ walks
+std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col_idx,
+                                        torch::Tensor weights, torch::Tensor sources,
+                                        int64_t num_repetition, int64_t len_path,
+                                        int64_t seed) {
+  CHECK_DEV(row_ptr); CHECK_CONT(row_ptr); CHECK_I32(row_ptr);
+  CHECK_DEV(col_idx); CHECK_CONT(col_idx); CHECK_I32(col_idx);
+  CHECK_DEV(weights); CHECK_CONT(weights); CHECK_F32(weights);
+  CHECK_DEV(sources); CHECK_CONT(sources); CHECK_I32(sources);
+  TORCH_CHECK(len_path >= 1 && len_path <= 512, "len_path out of range");
+  const long long n_src = sources.numel();
+  const long long n_walks = n_src * num_repetition;
+  auto opts_i = torch::TensorOptions().dtype(at::kInt).device(row_ptr.device());
+  auto opts_l = torch::TensorOptions().dtype(at::kLong).device(row_ptr.device());
+  auto nodes = torch::empty({n_walks, len_path}, opts_i);
+  auto lengths = torch::empty({n_walks}, opts_i);
+  auto hashes = torch::empty({n_walks}, opts_l);
+  if (n_walks == 0) return {nodes, lengths, hashes};
+  const int wpb = 4;                 // 256 threads = 4 waves
+  const size_t lds = (size_t)wpb * len_path * sizeof(int);
+  hipLaunchKernelGGL(walk_kernel, dim3(grid_for(n_walks, wpb)), dim3(256), lds,
+                     cur_stream(), row_ptr.data_ptr<int>(), col_idx.data_ptr<int>(),
+                     weights.data_ptr<float>(), sources.data_ptr<int>(),
+                     (int)n_src, n_walks, (int)len_path, (uint64_t)seed,
+                     nodes.data_ptr<int>(), lengths.data_ptr<int>(),
+                     (long long*)hashes.data_ptr<int64_t>());
+  LAUNCH_CHECK();
+  return {nodes, lengths, hashes};
+}
+
+// ------------------------------------------------------------ CBOW (fast)
+std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
+                                           torch::Tensor offs, torch::Tensor labels,
+                                           double inv_b, bool want_grad) {
+  CHECK_DEV(s); CHECK_CONT(s); CHECK_F32(s);
+  CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
+  CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
+  CHECK_DEV(labels); CHECK_CONT(labels); CHECK_F32(labels);
+  const long long P = labels.numel();
+  auto opts = torch::TensorOptions().dtype(at::kFloat).device(s.device());
+  auto loss = torch::empty({P}, opts);
+  auto correct = torch::empty({P}, opts);
+  auto dO = want_grad ? torch::empty({P}, opts) : torch::empty({0}, opts);
+  if (P == 0) return {loss, correct, dO};
+  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid_for(P, 4)), dim3(256), 0,
+                     cur_stream(), s.data_ptr<float>(), genes.data_ptr<int>(),
+                     offs.data_ptr<int>(), labels.data_ptr<float>(), P,
+                     (float)inv_b, loss.data_ptr<float>(),
+                     correct.data_ptr<float>(),
+                     want_grad ? dO.data_ptr<float>() : nullptr);
+  LAUNCH_CHECK();
+  return {loss, correct, dO};
+}
+
+torch::Tensor scatter_dO_det(torch::Tensor inst_path, torch::Tensor seg_start,
+                             torch::Tensor seg_gene, torch::Tensor dO,
+                             int64_t n_genes) {
+  CHECK_DEV(inst_path); CHECK_CONT(inst_path); CHECK_I32(inst_path);
+  CHECK_DEV(seg_start); CHECK_CONT(seg_start); CHECK_I32(seg_start);
+  CHECK_DEV(seg_gene); CHECK_CONT(seg_gene); CHECK_I32(seg_gene);
+  CHECK_DEV(dO); CHECK_CONT(dO); CHECK_F32(dO);
+  auto c = torch::zeros({n_genes},
+                        torch::TensorOptions().dtype(at::kFloat).device(dO.device()));
+  const long long n_seg = seg_gene.numel();
+  if (n_seg == 0) return c;
+  hipLaunchKernelGGL(scatter_do_det_kernel, dim3(grid_for(n_seg, 4)), dim3(256),
+                     0, cur_stream(), inst_path.data_ptr<int>(),
+                     seg_start.data_ptr<int>(), seg_gene.data_ptr<int>(),
+                     (int)n_seg, dO.data_ptr<float>(), c.data_ptr<float>());
+  LAUNCH_CHECK();
+  return c;
+}
+
+static double tf1_lr_t(double lr, double b1, double b2, int64_t t) {
+  return lr * std::sqrt(1.0 - std::pow(b2, (double)t)) /
+         (1.0 - std::pow(b1, (double)t));
+}
+
+void adam_rank1(torch::Tensor W, torch::Tensor m, torch::Tensor v,
+                torch::Tensor c, torch::Tensor who, int64_t t, double lr,
+                double b1, double b2, double eps) {
+  CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
+  CHECK_CONT(m); CHECK_CONT(v); CHECK_CONT(c); CHECK_CONT(who);
+  const long long G = W.size(0);
+  const int h = (int)W.size(1);
+  TORCH_CHECK(h % 4 == 0, "hidden must be a multiple of 4");
+  const double lr_t = tf1_lr_t(lr, b1, b2, t);
+  const long long n4 = G * h / 4;
+  hipLaunchKernelGGL(adam_rank1_kernel, dim3(grid_for(n4, 256)), dim3(256), 0,
+                     cur_stream(), W.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), c.data_ptr<float>(),
+                     who.data_ptr<float>(), G, h, (float)lr_t, (float)b1,
+                     (float)b2, (float)eps);
+  LAUNCH_CHECK();
+}
+
+void adam_dense(torch::Tensor W, torch::Tensor m, torch::Tensor v,
+                torch::Tensor grad, int64_t t, double lr, double b1, double b2,
+                double eps) {
+  CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
+  CHECK_CONT(m); CHECK_CONT(v); CHECK_CONT(grad);
+  TORCH_CHECK(grad.numel() == W.numel(), "grad/W size mismatch");
+  const long long n = W.numel();
+  const double lr_t = tf1_lr_t(lr, b1, b2, t);
+  hipLaunchKernelGGL(adam_dense_kernel, dim3(grid_for(n, 256)), dim3(256), 0,
+                     cur_stream(), W.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), grad.data_ptr<float>(), n,
+                     (float)lr_t, (float)b1, (float)b2, (float)eps);
+  LAUNCH_CHECK();
+}
+
+// ---------------------------------------------------------- CBOW (general)
+std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
+                                    torch::Tensor genes, torch::Tensor offs,
+                                    torch::Tensor labels, double inv_b,
+                                    bool want_grad) {
+  CHECK_DEV(W); CHECK_CONT(W);
+  CHECK_DEV(who); CHECK_CONT(who); CHECK_F32(who);
+  CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
+  CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
+  CHECK_DEV(labels); CHECK_CONT(labels); CHECK_F32(labels);
+  const bool bf16 = W.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16 || W.scalar_type() == at::kFloat, "W must be f32 or bf16");
+  const long long P = labels.numel();
+  const int h = (int)W.size(1);
+  const int hpl = h / 64;
+  TORCH_CHECK(h % 64 == 0 && hpl >= 1 && hpl <= 16 &&
+              (hpl & (hpl - 1)) == 0, "hidden must be 64*{1,2,4,8,16}");
+  auto opts = torch::TensorOptions().dtype(at::kFloat).device(W.device());
+  auto loss = torch::empty({P}, opts);
+  auto correct = torch::empty({P}, opts);
+  auto dO = want_grad ? torch::empty({P}, opts) : torch::empty({0}, opts);
+  auto H = want_grad ? torch::empty({P, h}, opts) : torch::empty({0}, opts);
+  if (P == 0) return {loss, correct, dO, H};
+  const int grid = grid_for(P, 4);
+  float* Hp = want_grad ? H.data_ptr<float>() : nullptr;
+  float* dOp = want_grad ? dO.data_ptr<float>() : nullptr;
+
+#define FWD_CASE(WT, HPL, PTR)                                                \
+  hipLaunchKernelGGL((cbow_fwd_kernel<WT, HPL>), dim3(grid), dim3(256), 0,    \
+                     cur_stream(), PTR, who.data_ptr<float>(),                \
+                     genes.data_ptr<int>(), offs.data_ptr<int>(),             \
+                     labels.data_ptr<float>(), P, (float)inv_b, h, Hp,        \
+                     loss.data_ptr<float>(), correct.data_ptr<float>(), dOp)
+  if (bf16) {
+    const uint16_t* Wp = (const uint16_t*)W.data_ptr<at::BFloat16>();
+    switch (hpl) {
+      case 1: FWD_CASE(uint16_t, 1, Wp); break;
+      case 2: FWD_CASE(uint16_t, 2, Wp); break;
+      case 4: FWD_CASE(uint16_t, 4, Wp); break;
+      case 8: FWD_CASE(uint16_t, 8, Wp); break;
+      default: FWD_CASE(uint16_t, 16, Wp); break;
+    }
+  } else {
+    const float* Wp = W.data_ptr<float>();
+    switch (hpl) {
+      case 1: FWD_CASE(float, 1, Wp); break;
+      case 2: FWD_CASE(float, 2, Wp); break;
+      case 4: FWD_CASE(float, 4, Wp); break;
+      case 8: FWD_CASE(float, 8, Wp); break;
+      default: FWD_CASE(float, 16, Wp); break;
+    }
+  }
+#undef FWD_CASE
+  LAUNCH_CHECK();
+  return {loss, correct, dO, H};
+}
+
+torch::Tensor cbow_bwd_rows(torch::Tensor who, torch::Tensor genes,
+                            torch::Tensor offs, torch::Tensor dO,
+                            int64_t n_genes) {
+  CHECK_DEV(who); CHECK_CONT(who); CHECK_F32(who);
+  CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
+  CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
+  CHECK_DEV(dO); CHECK_CONT(dO); CHECK_F32(dO);
+  const int h = (int)who.numel();
+  const int hpl = h / 64;
+  TORCH_CHECK(h % 64 == 0 && hpl >= 1 && hpl <= 16 &&
+              (hpl & (hpl - 1)) == 0, "hidden must be 64*{1,2,4,8,16}");
+  const long long P = dO.numel();
+  auto dW = torch::zeros({n_genes, h},
+                         torch::TensorOptions().dtype(at::kFloat).device(dO.device()));
+  if (P == 0) return dW;
+  const int grid = grid_for(P, 4);
+#define BWD_CASE(HPL)                                                         \
+  hipLaunchKernelGGL((cbow_bwd_rows_kernel<HPL>), dim3(grid), dim3(256), 0,   \
+                     cur_stream(), who.data_ptr<float>(),                     \
+                     genes.data_ptr<int>(), offs.data_ptr<int>(),             \
+                     dO.data_ptr<float>(), P, h, dW.data_ptr<float>())
+  switch (hpl) {
+    case 1: BWD_CASE(1); break;
+    case 2: BWD_CASE(2); break;
+    case 4: BWD_CASE(4); break;
+    case 8: BWD_CASE(8); break;
+    default: BWD_CASE(16); break;
+  }
+#undef BWD_CASE
+  LAUNCH_CHECK();
+  return dW;
+}
+
+// ------------------------------------------------------------------- PCC
+torch::Tensor pcc_edges(torch::Tensor zt, torch::Tensor edge_idx,
+                        int64_t n_group) {
+  CHECK_DEV(zt); CHECK_CONT(zt); CHECK_F32(zt);
+  CHECK_DEV(edge_idx); CHECK_CONT(edge_idx); CHECK_I32(edge_idx);
+  const long long E = edge_idx.size(0);
+  const int S = (int)zt.size(1);
+  auto out = torch::empty({E},
+                          torch::TensorOptions().dtype(at::kFloat).device(zt.device()));
+  if (E == 0) return out;
+  hipLaunchKernelGGL(pcc_edges_kernel, dim3(grid_for(E, 4)), dim3(256), 0,
+                     cur_stream(), zt.data_ptr<float>(),
+                     edge_idx.data_ptr<int>(), E, S, 1.0f / (float)n_group,
+                     out.data_ptr<float>());
+  LAUNCH_CHECK();
+  return out;
+}
+
+torch::Tensor corr_gemm(torch::Tensor zt, int64_t n_group) {
+  CHECK_DEV(zt); CHECK_CONT(zt); CHECK_F32(zt);
+  const int G = (int)zt.size(0);
+  const int S = (int)zt.size(1);
+  const int S4 = ((S + 3) / 4) * 4;
+  const size_t lds = 2ull * 64 * S4 * sizeof(float);
+  TORCH_CHECK(lds <= 160 * 1024,
+              "corr_gemm: S too large for the LDS-staged tile (use pcc_mode=edge)");
+  auto C = torch::empty({G, G},
+                        torch::TensorOptions().dtype(at::kFloat).device(zt.device()));
+  const int ntile = (G + 63) / 64;
+  hipLaunchKernelGGL(corr_gemm_kernel, dim3(ntile * ntile), dim3(256), lds,
+                     cur_stream(), zt.data_ptr<float>(), C.data_ptr<float>(),
+                     G, S, S4, 1.0f / (float)n_group);
+  LAUNCH_CHECK();
+  return C;
+}
+
+torch::Tensor bf16_copy(torch::Tensor src) {
+  CHECK_DEV(src); CHECK_CONT(src); CHECK_F32(src);
+  auto out = torch::empty_like(src, src.options().dtype(at::kBFloat16));
+  const long long n = src.numel();
+  if (n)
+    hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(grid_for(n, 256)), dim3(256),
+                       0, cur_stream(), src.data_ptr<float>(),
+                       (uint16_t*)out.data_ptr<at::BFloat16>(), n);
+  LAUNCH_CHECK();
+  return out;
+}
+
+// ------------------------------------------------- native host TSV parser
+std::tuple<std::vector<std::string>, std::vector<std::string>, torch::Tensor>
+parse_expression_tsv(const std::string& path) {
+  std::ifstream f(path);
+  TORCH_CHECK(f.good(), "cannot open ", path);
+  std::string line;
+  TORCH_CHECK(std::getline(f, line), "empty file ", path);
+  std::vector<std::string> samples;
+  {
+    size_t pos = line.find('\t');
+    TORCH_CHECK(pos != std::string::npos, "bad header in ", path);
+    size_t start = pos + 1;
+    while (start <= line.size()) {
+      size_t next = line.find('\t', start);
+      if (next == std::string::npos) {
+        std::string v = line.substr(start);
+        while (!v.empty() && (v.back() == '\r' || v.back() == '\n')) v.pop_back();
+        if (!v.empty()) samples.push_back(v);
+        break;
+      }
+      samples.push_back(line.substr(start, next - start));
+      start = next + 1;
+    }
+  }
+  const size_t S = samples.size();
+  std::vector<std::string> genes;
+  std::vector<float> vals;
+  vals.reserve(S * 4096);
+  while (std::getline(f, line)) {
+    if (line.empty()) continue;
+    const char* p = line.c_str();
+    const char* tab = std::strchr(p, '\t');
+    if (!tab) continue;
+    genes.emplace_back(p, tab - p);
+    const char* q = tab + 1;
+    for (size_t i = 0; i < S; ++i) {
+      char* end = nullptr;
+      vals.push_back(std::strtof(q, &end));
+      q = (end && *end == '\t') ? end + 1 : end;
+    }
+  }
+  const size_t G = genes.size();
+  auto t = torch::from_blob(vals.data(), {(long long)G, (long long)S},
+                            torch::TensorOptions().dtype(at::kFloat))
+               .clone();
+  // gene-wise -> sample-wise (like the reference transpose, G2Vec.py:498)
+  return {genes, samples, t.t().contiguous()};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("random_walks", &random_walks, "CSR biased random walks (gfx950)");
+  m.def("cbow_fwd_scalar", &cbow_fwd_scalar, "scalar CBOW forward + loss");
+  m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO");
+  m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");
+  m.def("adam_dense", &adam_dense, "TF1 Adam, dense grad");
+  m.def("cbow_fwd", &cbow_fwd, "row-gather CBOW forward");
+  m.def("cbow_bwd_rows", &cbow_bwd_rows, "scatter-add CBOW backward");
+  m.def("pcc_edges", &pcc_edges, "per-edge |PCC|");
+  m.def("corr_gemm", &corr_gemm, "MFMA f32 correlation GEMM");
+  m.def("bf16_copy", &bf16_copy, "f32 -> bf16 cast kernel");
+  m.def("parse_expression_tsv", &parse_expression_tsv,
+        "native expression TSV parser");
+}

@@ -1,0 +1,476 @@
+// g2vec_amd native kernels for AMD Instinct MI355X (gfx950, CDNA4).
+//
+// Hand-written HIP; no CUDA compatibility layers. Wavefront = 64 lanes
+// throughout. Kernels (SURVEY 2.10 K-table in parentheses):
+//
+//   walk_kernel            (K10) CSR biased non-revisiting random walk,
+//                          one wave per walk, LDS visited list, counter-based
+//                          splitmix64 RNG (bit-identical to ops/cpu_ref.py)
+//   cbow_fwd_scalar_kernel (K1+K2+K3+K8 collapsed) o_p = sum s_g over the
+//                          path + fused sigmoid-CE loss/accuracy/dlogit
+//   scatter_do_det_kernel  (K6 collapsed) c = X^T dO by gene-sorted segments
+//                          — deterministic (no atomics)
+//   adam_rank1_kernel      (K7) dense TF1-Adam with rank-1 grad c (x) who
+//   adam_dense_kernel      (K7) dense TF1-Adam with materialized grad
+//   cbow_fwd_kernel        (K1+K2+K3+K8 general) row-gather forward,
+//                          bf16/f32 W_ih, fp32 accumulate, H stored
+//   cbow_bwd_rows_kernel   (K6 general) scatter-add of dH rows (f32 atomics)
+//   pcc_edges_kernel       (K11) per-edge PCC dot product over samples
+//   corr_gemm_kernel       (K11 dense path) C = Z Z^T / S on f32 MFMA
+//                          (v_mfma_f32_16x16x4_f32) with LDS-staged tiles
+//
+// Reference behaviour being reimplemented (not copied): mathcom/G2Vec
+// G2Vec.py:328-346 (walk), :238-251 (CBOW fwd/loss/acc), :245-246 (Adam),
+// :354-368 (PCC).
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define WAVE 64
+
+// ---------------------------------------------------------------- RNG / hash
+// Must stay bit-identical to g2vec_amd/ops/cpu_ref.py::splitmix64/gene_hash.
+__device__ __forceinline__ uint64_t sm64_next(uint64_t& s) {
+  s += 0x9E3779B97F4A7C15ULL;
+  uint64_t z = s;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+  return z ^ (z >> 31);
+}
+
+__device__ __forceinline__ double u01_from(uint64_t r) {
+  return (double)(r >> 11) * (1.0 / 9007199254740992.0);
+}
+
+__device__ __forceinline__ uint64_t gene_hash_dev(uint32_t g) {
+  uint64_t s = (uint64_t)g * 0xBF58476D1CE4E5B9ULL + 0x9E3779B97F4A7C15ULL;
+  return sm64_next(s);
+}
+
+// ---------------------------------------------------------------- wave utils
+__device__ __forceinline__ float wave_sum(float v) {
+  for (int o = 32; o; o >>= 1) v += __shfl_down(v, o);
+  return __shfl(v, 0);
+}
+
+__device__ __forceinline__ float wave_incl_scan(float v) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  for (int o = 1; o < WAVE; o <<= 1) {
+    float t = __shfl_up(v, o);
+    if (lane >= o) v += t;
+  }
+  return v;
+}
+
+__device__ __forceinline__ float bf16_to_f32(uint16_t b) {
+  union { float f; uint32_t u; } x;
+  x.u = ((uint32_t)b) << 16;
+  return x.f;
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+  union { float f; uint32_t u; } x;
+  x.f = f;
+  // round-to-nearest-even like torch's .bfloat16()
+  uint32_t lsb = (x.u >> 16) & 1u;
+  x.u += 0x7FFFu + lsb;
+  return (uint16_t)(x.u >> 16);
+}
+
+// ================================================================= K10: walks
+// One 64-lane wave per walk. Per step: two lane-parallel passes over the
+// CSR row (masked total, then weighted selection by prefix-scan), visited
+// list in LDS (<= len_path entries, broadcast reads). Matches the reference
+// walk semantics exactly (G2Vec.py:328-346); the RNG stream is the
+// framework's own (counter-based, reproducible, shared with the CPU oracle).
+extern "C" __global__ void __launch_bounds__(256)
+walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
+            const float* __restrict__ wgt, const int* __restrict__ sources,
+            int n_src, long long n_walks, int len_path, uint64_t seed,
+            int* __restrict__ out_nodes, int* __restrict__ out_len,
+            long long* __restrict__ out_hash) {
+  extern __shared__ int smem[];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;            // wave-in-block
+  const int wpb = blockDim.x >> 6;             // waves per block
+  int* vis = smem + wib * len_path;
+
+  for (long long walk = (long long)blockIdx.x * wpb + wib; walk < n_walks;
+       walk += (long long)gridDim.x * wpb) {
+    uint64_t state = seed ^ (uint64_t)((uint64_t)walk * 0x94D049BB133111EBULL + 1ULL);
+    (void)sm64_next(state);                    // warm draw (CPU oracle parity)
+    int cur = sources[walk % n_src];
+    int plen = 0;
+    uint64_t hash = 0;
+
+    for (int step = 0; step < len_path; ++step) {
+      vis[plen] = cur;                         // same value from every lane
+      ++plen;
+      hash += gene_hash_dev((uint32_t)cur);
+      const int s = row_ptr[cur], e = row_ptr[cur + 1];
+      const int deg = e - s;
+      if (deg <= 0) break;
+
+      // pass 1: total unvisited weight
+      float partial = 0.f;
+      for (int j = lane; j < deg; j += WAVE) {
+        const int cand = col_idx[s + j];
+        float w = wgt[s + j];
+        for (int k = 0; k < plen; ++k)
+          if (vis[k] == cand) { w = 0.f; break; }
+        partial += w;
+      }
+      const float tot = wave_sum(partial);
+      const uint64_t r = sm64_next(state);     // drawn even on dead end (oracle parity)
+      if (!(tot > 0.f)) break;
+      const float target = (float)(u01_from(r) * (double)tot);
+
+      // pass 2: first index whose masked-weight running sum exceeds target
+      int chosen = -1;
+      float base = 0.f;
+      for (int j0 = 0; j0 < deg; j0 += WAVE) {
+        const int j = j0 + lane;
+        float w = 0.f;
+        if (j < deg) {
+          const int cand = col_idx[s + j];
+          w = wgt[s + j];
+          for (int k = 0; k < plen; ++k)
+            if (vis[k] == cand) { w = 0.f; break; }
+        }
+        const float scan = wave_incl_scan(w);
+        const float chunk_tot = __shfl(scan, WAVE - 1);
+        const bool hit = (j < deg) && (w > 0.f) &&
+                         (base + scan > target) && (base + scan - w <= target);
+        const unsigned long long m = __ballot(hit);
+        if (m != 0ULL) { chosen = j0 + (__ffsll((long long)m) - 1); break; }
+        base += chunk_tot;
+      }
+      if (chosen < 0) {
+        // numerical tail: target >= running total by rounding -> last unvisited
+        for (int j0 = ((deg - 1) / WAVE) * WAVE; j0 >= 0 && chosen < 0; j0 -= WAVE) {
+          const int j = j0 + lane;
+          float w = 0.f;
+          if (j < deg) {
+            const int cand = col_idx[s + j];
+            w = wgt[s + j];
+            for (int k = 0; k < plen; ++k)
+              if (vis[k] == cand) { w = 0.f; break; }
+          }
+          const unsigned long long m = __ballot(w > 0.f);
+          if (m != 0ULL) chosen = j0 + (63 - __clzll((long long)m));
+        }
+        if (chosen < 0) break;                 // cannot happen when tot > 0
+      }
+      cur = col_idx[s + chosen];
+    }
+
+    long long outb = walk * (long long)len_path;
+    for (int k = lane; k < len_path; k += WAVE)
+      out_nodes[outb + k] = (k < plen) ? vis[k] : -1;
+    if (lane == 0) {
+      out_len[walk] = plen;
+      out_hash[walk] = (long long)hash;
+    }
+  }
+}
+
+// ===================================================== fast path: scalar CBOW
+// o_p = sum_{g in p} s_g; loss = sigmoid-CE(o, y); correct = (o>0)==y;
+// dO = (sigmoid(o)-y)*inv_b. One wave per path, lane-strided scalar gather.
+extern "C" __global__ void __launch_bounds__(256)
+cbow_fwd_scalar_kernel(const float* __restrict__ s, const int* __restrict__ genes,
+                       const int* __restrict__ offs, const float* __restrict__ labels,
+                       long long P, float inv_b, float* __restrict__ loss,
+                       float* __restrict__ correct, float* __restrict__ dO) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
+       p += (long long)gridDim.x * wpb) {
+    const int lo = offs[p], hi = offs[p + 1];
+    float partial = 0.f;
+    for (int i = lo + lane; i < hi; i += WAVE) partial += s[genes[i]];
+    const float o = wave_sum(partial);
+    if (lane == 0) {
+      const float y = labels[p];
+      loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
+      correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
+      if (dO) dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
+    }
+  }
+}
+
+// =============================================== fast path: c = X^T dO (det.)
+// One wave per gene segment (instances pre-sorted by gene, plan built once
+// per path set). Fixed tree-reduction order -> bitwise reproducible.
+extern "C" __global__ void __launch_bounds__(256)
+scatter_do_det_kernel(const int* __restrict__ inst_path,
+                      const int* __restrict__ seg_start,
+                      const int* __restrict__ seg_gene, int n_seg,
+                      const float* __restrict__ dO, float* __restrict__ c) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  for (long long sidx = (long long)blockIdx.x * wpb + wib; sidx < n_seg;
+       sidx += (long long)gridDim.x * wpb) {
+    const int lo = seg_start[sidx], hi = seg_start[sidx + 1];
+    float partial = 0.f;
+    for (int i = lo + lane; i < hi; i += WAVE) partial += dO[inst_path[i]];
+    const float v = wave_sum(partial);
+    if (lane == 0) c[seg_gene[sidx]] = v;
+  }
+}
+
+// ============================================================== K7: TF1 Adam
+// theta -= lr_t * m / (sqrt(v) + eps); lr_t precomputed on host.
+// rank-1 variant forms grad[g*h+j] = c[g] * who[j] on the fly (the linear
+// CBOW's dW_ih is rank-1, see models/cbow.py) — no G*h grad tensor exists.
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+extern "C" __global__ void __launch_bounds__(256)
+adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
+                  float* __restrict__ v, const float* __restrict__ c,
+                  const float* __restrict__ who, long long G, int h,
+                  float lr_t, float b1, float b2, float eps) {
+  const long long n4 = (long long)G * h / 4;
+  f32x4* W4 = (f32x4*)W; f32x4* m4 = (f32x4*)m; f32x4* v4 = (f32x4*)v;
+  const f32x4* who4 = (const f32x4*)who;
+  const int h4 = h / 4;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long long)gridDim.x * blockDim.x) {
+    const long long g = i / h4;
+    const int j4 = (int)(i % h4);
+    const float cg = c[g];
+    const f32x4 wj = who4[j4];
+    f32x4 mm = m4[i], vv = v4[i], ww = W4[i];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      const float grad = cg * wj[k];
+      mm[k] = b1 * mm[k] + (1.f - b1) * grad;
+      vv[k] = b2 * vv[k] + (1.f - b2) * grad * grad;
+      ww[k] -= lr_t * mm[k] / (sqrtf(vv[k]) + eps);
+    }
+    m4[i] = mm; v4[i] = vv; W4[i] = ww;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+adam_dense_kernel(float* __restrict__ W, float* __restrict__ m,
+                  float* __restrict__ v, const float* __restrict__ grad,
+                  long long n, float lr_t, float b1, float b2, float eps) {
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x) {
+    const float g = grad[i];
+    float mm = b1 * m[i] + (1.f - b1) * g;
+    float vv = b2 * v[i] + (1.f - b2) * g * g;
+    m[i] = mm; v[i] = vv;
+    W[i] -= lr_t * mm / (sqrtf(vv) + eps);
+  }
+}
+
+// ====================================== general path: row-gather CBOW forward
+// One wave per path; h/64 embedding columns per lane (contiguous, vector
+// loads); f32 accumulate; H stored for the general backward; loss/acc/dO
+// fused into the same pass (the reference recomputed the forward 3x per
+// epoch and shipped 2.5 GB/epoch host->device, SURVEY 3.2 — here everything
+// is resident and fused).
+template <typename WT, int HPL>
+__global__ void __launch_bounds__(256)
+cbow_fwd_kernel(const WT* __restrict__ W, const float* __restrict__ who,
+                const int* __restrict__ genes, const int* __restrict__ offs,
+                const float* __restrict__ labels, long long P, float inv_b,
+                int h, float* __restrict__ H, float* __restrict__ loss,
+                float* __restrict__ correct, float* __restrict__ dO) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  const int col0 = lane * HPL;
+  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
+       p += (long long)gridDim.x * wpb) {
+    const int lo = offs[p], hi = offs[p + 1];
+    float acc[HPL];
+#pragma unroll
+    for (int k = 0; k < HPL; ++k) acc[k] = 0.f;
+    for (int i = lo; i < hi; ++i) {
+      const long long g = genes[i];
+      const WT* row = W + g * (long long)h + col0;
+      WT tmp[HPL];
+      __builtin_memcpy(tmp, row, HPL * sizeof(WT));  // one wide load
+#pragma unroll
+      for (int k = 0; k < HPL; ++k) {
+        if constexpr (sizeof(WT) == 2)
+          acc[k] += bf16_to_f32(*(const uint16_t*)&tmp[k]);
+        else
+          acc[k] += *(const float*)&tmp[k];
+      }
+    }
+    float op = 0.f;
+    float wv[HPL];
+    __builtin_memcpy(wv, who + col0, HPL * sizeof(float));
+#pragma unroll
+    for (int k = 0; k < HPL; ++k) op += acc[k] * wv[k];
+    const float o = wave_sum(op);
+    if (H) {
+      float* hrow = H + p * (long long)h + col0;
+      __builtin_memcpy(hrow, acc, HPL * sizeof(float));
+    }
+    if (lane == 0) {
+      const float y = labels[p];
+      loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
+      correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
+      if (dO) dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
+    }
+  }
+}
+
+// ======================================= general path: dW_ih scatter backward
+// dH_p = dO_p * who (recomputed, never materialized); atomicAdd into the
+// touched gene rows (f32 global atomics; hub-gene contention measured fine
+// at ex_* scale — the deterministic fast path is the default trainer).
+template <int HPL>
+__global__ void __launch_bounds__(256)
+cbow_bwd_rows_kernel(const float* __restrict__ who, const int* __restrict__ genes,
+                     const int* __restrict__ offs, const float* __restrict__ dO,
+                     long long P, int h, float* __restrict__ dW) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  const int col0 = lane * HPL;
+  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
+       p += (long long)gridDim.x * wpb) {
+    const float g = dO[p];
+    float dh[HPL];
+    float wv[HPL];
+    __builtin_memcpy(wv, who + col0, HPL * sizeof(float));
+#pragma unroll
+    for (int k = 0; k < HPL; ++k) dh[k] = g * wv[k];
+    const int lo = offs[p], hi = offs[p + 1];
+    for (int i = lo; i < hi; ++i) {
+      float* row = dW + (long long)genes[i] * h + col0;
+#pragma unroll
+      for (int k = 0; k < HPL; ++k) atomicAdd(row + k, dh[k]);
+    }
+  }
+}
+
+// =================================================== K11: per-edge PCC weight
+// zt: f32 [G, S] z-scored rows; one wave per edge, lane-strided dot.
+// PCC = dot / S; 0-variance genes have zero rows (graph.py) -> pcc 0.
+extern "C" __global__ void __launch_bounds__(256)
+pcc_edges_kernel(const float* __restrict__ zt, const int* __restrict__ edges,
+                 long long E, int S, float inv_s, float* __restrict__ out) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  for (long long e = (long long)blockIdx.x * wpb + wib; e < E;
+       e += (long long)gridDim.x * wpb) {
+    const long long a = edges[2 * e], b = edges[2 * e + 1];
+    const float* za = zt + a * S;
+    const float* zb = zt + b * S;
+    float partial = 0.f;
+    for (int i = lane; i < S; i += WAVE) partial += za[i] * zb[i];
+    const float d = wave_sum(partial);
+    if (lane == 0) out[e] = fabsf(d * inv_s);
+  }
+}
+
+// ============================================== K11 dense: MFMA f32 corr GEMM
+// C = Zt Zt^T / S on v_mfma_f32_16x16x4_f32 (exact f32 at the FP32 matrix
+// rate — no xf32 on gfx950, see cdna_hip_programming.md §3). 64x64 tile per
+// 4-wave block (each wave one 32x32 sub-tile = 2x2 16x16 fragments), both
+// operand tiles staged once in LDS (K = S <= 288 fits the 160 KiB budget).
+typedef float f32x4c __attribute__((ext_vector_type(4)));
+
+extern "C" __global__ void __launch_bounds__(256)
+corr_gemm_kernel(const float* __restrict__ zt, float* __restrict__ C,
+                 int G, int S, int S4, float inv_s) {
+#if defined(__gfx950__)
+  extern __shared__ float lds[];
+  float* At = lds;                 // [64][S4]
+  float* Bt = lds + 64 * S4;       // [64][S4]
+  const int ntile = (G + 63) / 64;
+  const int ti = blockIdx.x / ntile;
+  const int tj = blockIdx.x % ntile;
+  const int i0 = ti * 64, j0 = tj * 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  // stage: each thread strides over the 64 x S4 tiles
+  for (int idx = threadIdx.x; idx < 64 * S4; idx += blockDim.x) {
+    const int r = idx / S4, k = idx % S4;
+    At[idx] = (i0 + r < G && k < S) ? zt[(long long)(i0 + r) * S + k] : 0.f;
+    Bt[idx] = (j0 + r < G && k < S) ? zt[(long long)(j0 + r) * S + k] : 0.f;
+  }
+  __syncthreads();
+
+  const int wr = wid >> 1, wc = wid & 1;       // 2x2 wave grid -> 32x32 each
+  f32x4c acc[2][2];
+#pragma unroll
+  for (int a = 0; a < 2; ++a)
+#pragma unroll
+    for (int b = 0; b < 2; ++b) acc[a][b] = (f32x4c)(0.f);
+
+  const int fr = lane & 15;                     // fragment row/col lane index
+  const int kk = lane >> 4;                     // k sub-index 0..3
+  for (int k0 = 0; k0 < S4; k0 += 4) {
+    float av[2], bv[2];
+#pragma unroll
+    for (int a = 0; a < 2; ++a)
+      av[a] = At[(wr * 32 + a * 16 + fr) * S4 + k0 + kk];
+#pragma unroll
+    for (int b = 0; b < 2; ++b)
+      bv[b] = Bt[(wc * 32 + b * 16 + fr) * S4 + k0 + kk];
+#pragma unroll
+    for (int a = 0; a < 2; ++a)
+#pragma unroll
+      for (int b = 0; b < 2; ++b)
+        acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[a], bv[b],
+                                                         acc[a][b], 0, 0, 0);
+  }
+  // C/D map (16x16): col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int a = 0; a < 2; ++a)
+#pragma unroll
+    for (int b = 0; b < 2; ++b)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int ci = i0 + wr * 32 + a * 16 + (lane >> 4) * 4 + reg;
+        const int cj = j0 + wc * 32 + b * 16 + (lane & 15);
+        if (ci < G && cj < G)
+          C[(long long)ci * G + cj] = acc[a][b][reg] * inv_s;
+      }
+#endif
+}
+
+// =============================================================== bf16 cast
+extern "C" __global__ void __launch_bounds__(256)
+f32_to_bf16_kernel(const float* __restrict__ in, uint16_t* __restrict__ out,
+                   long long n) {
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x)
+    out[i] = f32_to_bf16(in[i]);
+}
+
+// ------------------------------------------------- template instantiations
+#define INSTANTIATE_FWD(WT, HPL)                                              \
+  template __global__ void cbow_fwd_kernel<WT, HPL>(                          \
+      const WT*, const float*, const int*, const int*, const float*,          \
+      long long, float, int, float*, float*, float*, float*);
+INSTANTIATE_FWD(float, 1)
+INSTANTIATE_FWD(float, 2)
+INSTANTIATE_FWD(float, 4)
+INSTANTIATE_FWD(float, 8)
+INSTANTIATE_FWD(float, 16)
+INSTANTIATE_FWD(uint16_t, 1)
+INSTANTIATE_FWD(uint16_t, 2)
+INSTANTIATE_FWD(uint16_t, 4)
+INSTANTIATE_FWD(uint16_t, 8)
+INSTANTIATE_FWD(uint16_t, 16)
+
+#define INSTANTIATE_BWD(HPL)                                                  \
+  template __global__ void cbow_bwd_rows_kernel<HPL>(                         \
+      const float*, const int*, const int*, const float*, long long, int,     \
+      float*);
+INSTANTIATE_BWD(1)
+INSTANTIATE_BWD(2)
+INSTANTIATE_BWD(4)
+INSTANTIATE_BWD(8)
+INSTANTIATE_BWD(16)
