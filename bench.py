@@ -1,0 +1,193 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — the BASELINE.json headline on the ex_*-shaped config.
+
+Metric: CBOW trained paths/sec (+ wall-clock to val-ACC >= 0.88, reported in
+`config`) on the bundled-dataset shape: 7,523 common genes / ~216k restricted
+edges / 135 samples / lenPath 80 / 10 repetitions / hidden 128 — synthetic
+data of that shape (the reference's ex_EXPRESSION.txt is not distributable)
+and random-init weights.
+
+One timed "step" = one full reference training epoch: the full-batch
+optimizer pass at W_t (forward + rank-1 backward + grad all-reduce + dense
+TF1-Adam) plus the post-update train-ACC and val-ACC evaluations — exactly
+the per-epoch work of the reference (G2Vec.py:262-267), which its ~2.2
+s/epoch baseline also includes.
+
+Scaling is WEAK: each rank generates and trains its own ex_*-shaped path
+shard (global batch grows with N); gradients are summed over ranks with a
+single fused RCCL all-reduce (the rank-1 backward makes the whole grad
+message G+h+2 floats).
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+Launched multi-GPU by the driver via torch.distributed.run (one rank/GPU).
+Rank 0 prints ONE JSON line on stdout; progress goes to stderr.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+from g2vec_amd.config import G2VecConfig
+from g2vec_amd.graph import build_group_graph
+from g2vec_amd.models.cbow import CbowTrainer
+from g2vec_amd.parallel.dist import init_dist
+from g2vec_amd.paths import integrate_pathsets
+from g2vec_amd.utils import synth
+from g2vec_amd.walks import generate_walks
+
+BASELINE_PATHS_PER_SEC = 16500.0   # BASELINE.md derived reference number
+
+
+def log(*a):
+    print(*a, file=sys.stderr, flush=True)
+
+
+def build_dataset(seed: int):
+    """In-memory ex_*-shaped dataset (one per rank; weak scaling)."""
+    n_genes, n_extra = 7523, 2381
+    rng = np.random.default_rng(seed)
+    n_net = n_genes + n_extra
+    module = np.full(n_net, -1, dtype=np.int64)
+    live = rng.choice(n_genes, size=int(n_genes * 0.5), replace=False)
+    module[live] = rng.integers(0, 16, size=live.size)
+    _, edge_idx, _ = synth.synth_network(n_net, 298799, 16, seed, module=module)
+    # restrict to the common genes (network extras have no expression)
+    keep = (edge_idx[:, 0] < n_genes) & (edge_idx[:, 1] < n_genes)
+    edge_idx = edge_idx[keep]
+    _, labels = synth.synth_clinical(135, 58, seed)
+    expr = synth.synth_expression(
+        [f"G{i}" for i in range(n_genes)], labels, module[:n_genes], seed)
+    return expr, np.asarray(labels), edge_idx.astype(np.int32), n_genes
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--hidden", type=int, default=128)
+    ap.add_argument("--len-path", type=int, default=80)
+    ap.add_argument("--reps", type=int, default=10)
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--trainer-path", choices=["fast", "general"],
+                    default="fast")
+    args = ap.parse_args()
+
+    ctx = init_dist("auto")
+    device = ctx.device
+    on_gpu = device.type == "cuda"
+    if on_gpu:
+        import g2vec_amd.ops as ops
+        assert ops.native_available(), \
+            "bench on GPU requires the native extension (build_ext --inplace)"
+    world = ctx.world
+    rank = ctx.rank
+    log(f"[bench] rank {rank}/{world} device {device}")
+
+    cfg = G2VecConfig(hidden=args.hidden, len_path=args.len_path,
+                      num_repetition=args.reps, epochs=500, seed=args.seed,
+                      device=str(device.type), trainer_path=args.trainer_path)
+
+    # ---- dataset + graphs + walks (input pipeline; measured, not the metric)
+    expr, labels, edge_idx, n_genes = build_dataset(args.seed + 1000 * rank)
+    expr_t = torch.from_numpy(expr).to(device)
+    labels_t = torch.from_numpy(labels).to(device)
+    edge_t = torch.from_numpy(edge_idx).to(device)
+
+    t0 = time.perf_counter()
+    walksets = []
+    for group in (0, 1):
+        g = build_group_graph(expr_t, labels_t, group, edge_t, n_genes)
+        walksets.append(generate_walks(g, cfg.len_path, cfg.num_repetition,
+                                       cfg.seed + rank, group))
+    if on_gpu:
+        torch.cuda.synchronize()
+    walk_s = time.perf_counter() - t0
+    n_walks = sum(int(w.nodes.shape[0]) for w in walksets)
+    ps, _freq, _nip = integrate_pathsets(walksets[0], walksets[1], n_genes)
+    log(f"[bench] rank {rank}: {n_walks} walks in {walk_s:.3f}s "
+        f"({n_walks / walk_s:.0f} walks/s), {ps.n_paths} paths after dedup")
+
+    # ---- convergence probe: epochs + wall to val-ACC >= 0.88 (untimed work
+    # for the throughput metric, but itself the secondary headline)
+    trainer = CbowTrainer(cfg, n_genes, device, ctx,
+                          log=(lambda *a, **k: None))
+    st = trainer.setup(ps, pre_sharded=True)
+    wall_to_acc = None
+    acc_val = 0.0
+    conv_t0 = time.perf_counter()
+    for ep in range(60):
+        _acc_tr, acc_val = trainer.run_epoch(st)
+        if acc_val >= 0.88:
+            if on_gpu:
+                torch.cuda.synchronize()
+            wall_to_acc = time.perf_counter() - conv_t0
+            break
+    log(f"[bench] rank {rank}: val-ACC {acc_val:.4f} "
+        f"(wall-to-0.88: {wall_to_acc})")
+
+    # ---- timed throughput region: fresh state, W warmup + K timed epochs
+    st = trainer.setup(ps, pre_sharded=True)
+    n_tr_global = trainer.n_tr_global
+    for _ in range(args.warmup):
+        trainer.run_epoch(st)
+    ctx.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.run_epoch(st)
+    if on_gpu:
+        torch.cuda.synchronize()
+    ctx.barrier()
+    elapsed = time.perf_counter() - t0
+    emax = torch.tensor([elapsed], dtype=torch.float64, device=device)
+    ctx.allreduce_max_(emax)
+    elapsed = float(emax.item())
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    value = n_tr_global * args.steps / elapsed
+
+    if ctx.is_primary:
+        out = {
+            "metric": "cbow_paths_per_sec",
+            "value": round(value, 1),
+            "unit": "paths/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / BASELINE_PATHS_PER_SEC, 2),
+            "dtype": "fp32",
+            "data": "synthetic (ex_* shape: 7523 genes/135 samples/216k edges; "
+                    "random-init weights)",
+            "config": {
+                "model": "g2vec-cbow",
+                "global_batch": n_tr_global,
+                "seq_len": args.len_path,
+                "parallelism": f"dp{world}",
+                "hidden": args.hidden,
+                "num_repetition": args.reps,
+                "trainer_path": args.trainer_path,
+                "val_acc": round(acc_val, 4),
+                "wall_to_val_acc_0.88_s": (round(wall_to_acc, 4)
+                                           if wall_to_acc else None),
+                "walks_per_sec": round(n_walks / walk_s, 1),
+                "step_includes": "full-batch fwd+bwd+allreduce+dense-Adam "
+                                 "+ post-update train/val ACC evals",
+            },
+        }
+        print(json.dumps(out), flush=True)
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

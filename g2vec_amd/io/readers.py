@@ -17,18 +17,19 @@ from typing import Dict, List, Set, Tuple
 import numpy as np
 
 
-def load_expression(path: str) -> dict:
+def load_expression(path: str, use_native: bool = True) -> dict:
     """Returns {'sample': [S] str array, 'expr': f32 [S, G], 'gene': [G] str array}."""
-    try:
-        from g2vec_amd import _C  # native fast path
-        genes, samples, expr = _C.parse_expression_tsv(path)
-        return {
-            "sample": np.array(samples),
-            "expr": expr.numpy().astype(np.float32, copy=False),
-            "gene": np.array(genes),
-        }
-    except ImportError:
-        pass
+    if use_native:
+        try:
+            from g2vec_amd import _C  # native fast path
+            genes, samples, expr = _C.parse_expression_tsv(path)
+            return {
+                "sample": np.array(samples),
+                "expr": expr.numpy().astype(np.float32, copy=False),
+                "gene": np.array(genes),
+            }
+        except ImportError:
+            pass
     with open(path) as fin:
         header = fin.readline().rstrip("\n").split("\t")
         samples = header[1:]

@@ -80,23 +80,36 @@ class CbowTrainer:
                 who.copy_(_trunc_normal(who.shape, std, gen))
         return W.to(self.device), who.to(self.device)
 
-    def _split(self, ps: PathSet):
-        """Seeded shuffle + 80/20 split (G2Vec.py:219-226), then DP shard."""
+    def _split(self, ps: PathSet, pre_sharded: bool = False):
+        """Seeded shuffle + 80/20 split (G2Vec.py:219-226), then DP shard.
+
+        pre_sharded=True means `ps` is already this rank's local shard
+        (weak-scaling benches): split locally, count globally."""
         P = ps.n_paths
         gen = torch.Generator()
         if self.cfg.seed is not None:
-            gen.manual_seed(int(self.cfg.seed) + 12345)
+            gen.manual_seed(int(self.cfg.seed) + 12345 +
+                            (self.ctx.rank if pre_sharded else 0))
         perm = torch.randperm(P, generator=gen).to(self.device)
         pivot = int(P * 0.8)
         tr_idx, vl_idx = perm[:pivot], perm[pivot:]
-        self.n_tr_global = int(tr_idx.numel())
-        self.n_vl_global = int(vl_idx.numel())
-        tr_idx = tr_idx[self.ctx.shard_indices(tr_idx.numel(), self.device)]
-        vl_idx = vl_idx[self.ctx.shard_indices(vl_idx.numel(), self.device)]
+        if pre_sharded:
+            counts = torch.tensor([tr_idx.numel(), vl_idx.numel()],
+                                  dtype=torch.float64, device=self.device)
+            self.ctx.allreduce_(counts)
+            self.n_tr_global = int(counts[0].item())
+            self.n_vl_global = int(counts[1].item())
+        else:
+            self.n_tr_global = int(tr_idx.numel())
+            self.n_vl_global = int(vl_idx.numel())
+            tr_idx = tr_idx[self.ctx.shard_indices(tr_idx.numel(), self.device)]
+            vl_idx = vl_idx[self.ctx.shard_indices(vl_idx.numel(), self.device)]
         return subset(ps, tr_idx), subset(ps, vl_idx)
 
-    # ------------------------------------------------------------------ train
-    def train(self, ps: PathSet) -> TrainResult:
+    # ------------------------------------------------------------------ setup
+    def setup(self, ps: PathSet, pre_sharded: bool = False):
+        """Initialise weights/optimizer state and split the path set.
+        Returns the mutable training state used by run_epoch()."""
         cfg = self.cfg
         gen = None
         if cfg.seed is not None:
@@ -105,32 +118,54 @@ class CbowTrainer:
         W, who = self._init_weights(gen)
         self.ctx.broadcast_(W)      # C4: replicated params (all ranks identical)
         self.ctx.broadcast_(who)
-        tr, vl = self._split(ps)
-
-        mW = torch.zeros_like(W)
-        vW = torch.zeros_like(W)
-        mO = torch.zeros_like(who)
-        vO = torch.zeros_like(who)
-        W_keep = W.clone()
+        tr, vl = self._split(ps, pre_sharded)
 
         use_general = cfg.trainer_path == "general"
-        W16 = W.bfloat16() if (use_general and cfg.dtype == "bf16") else None
-        plan = None
+        st = type("TrainState", (), {})()
+        st.W, st.who = W, who
+        st.mW, st.vW = torch.zeros_like(W), torch.zeros_like(W)
+        st.mO, st.vO = torch.zeros_like(who), torch.zeros_like(who)
+        st.W_keep = W.clone()
+        st.tr, st.vl = tr, vl
+        st.W16 = W.bfloat16() if (use_general and cfg.dtype == "bf16") else None
+        st.plan = None
         if not use_general:
-            plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)
-
-        inv_b = 1.0 / max(self.n_tr_global, 1)
-        # minibatch boundaries over the local shard (full batch by default)
+            st.plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)
+        st.inv_b = 1.0 / max(self.n_tr_global, 1)
         P_loc = tr.n_paths
         bs = cfg.batch_size if cfg.batch_size > 0 else P_loc
-        batches = [(i, min(i + bs, P_loc)) for i in range(0, max(P_loc, 1), bs)]
+        st.batches = [(i, min(i + bs, P_loc)) for i in range(0, max(P_loc, 1), bs)]
+        st.t_adam = 0
+        return st
+
+    def run_epoch(self, st) -> tuple:
+        """One reference epoch: optimizer step(s) at W_t, then post-update
+        accuracy on both splits (G2Vec.py:262-267). Returns (acc_tr, acc_val)."""
+        cfg = self.cfg
+        for (lo, hi) in st.batches:
+            b_inv = st.inv_b if cfg.batch_size == 0 else 1.0 / (
+                (hi - lo) * self.ctx.world)
+            st.t_adam += 1
+            if cfg.trainer_path == "general":
+                self._step_general(st.W, st.W16, st.who, st.mW, st.vW, st.mO,
+                                   st.vO, st.tr, lo, hi, b_inv, st.t_adam)
+            else:
+                self._step_fast(st.W, st.who, st.mW, st.vW, st.mO, st.vO,
+                                st.tr, st.plan, lo, hi, b_inv, st.t_adam)
+        acc_tr = self._accuracy(st.W, st.W16, st.who, st.tr, self.n_tr_global)
+        acc_val = self._accuracy(st.W, st.W16, st.who, st.vl, self.n_vl_global)
+        return acc_tr, acc_val
+
+    # ------------------------------------------------------------------ train
+    def train(self, ps: PathSet, pre_sharded: bool = False) -> TrainResult:
+        cfg = self.cfg
+        st = self.setup(ps, pre_sharded)
 
         before_val, before_tr = -1.0, -1.0
         stop_epoch = -1
         acc_hist: List[float] = []
         epoch_times: List[float] = []
         wall_to_acc = None
-        t_adam = 0
         t0_all = time.perf_counter()
         display_step = 5
         blk_t0 = time.perf_counter()
@@ -139,20 +174,7 @@ class CbowTrainer:
         epochs_run = 0
         for epoch in range(cfg.epochs):
             ep_t0 = time.perf_counter()
-            for (lo, hi) in batches:
-                b_inv = inv_b if cfg.batch_size == 0 else 1.0 / (
-                    (hi - lo) * self.ctx.world)
-                t_adam += 1
-                if use_general:
-                    self._step_general(W, W16, who, mW, vW, mO, vO, tr, lo, hi,
-                                       b_inv, t_adam)
-                else:
-                    self._step_fast(W, who, mW, vW, mO, vO, tr, plan, lo, hi,
-                                    b_inv, t_adam)
-            # post-update accuracy on both splits (reference order,
-            # G2Vec.py:264-267) — one fused scalar forward per split
-            acc_tr = self._accuracy(W, W16, who, tr, self.n_tr_global)
-            acc_val = self._accuracy(W, W16, who, vl, self.n_vl_global)
+            acc_tr, acc_val = self.run_epoch(st)
             epochs_run = epoch + 1
             acc_hist.append(acc_val)
             epoch_times.append(time.perf_counter() - ep_t0)
@@ -172,10 +194,10 @@ class CbowTrainer:
                 acc_val, acc_tr = before_val, before_tr
                 break
             before_val, before_tr = acc_val, acc_tr
-            W_keep.copy_(W)         # keep-last-good snapshot (G2Vec.py:283)
+            st.W_keep.copy_(st.W)   # keep-last-good snapshot (G2Vec.py:283)
         self.log("    Optimization Finish")
 
-        return TrainResult(W_ih=W_keep, stop_epoch=stop_epoch,
+        return TrainResult(W_ih=st.W_keep, stop_epoch=stop_epoch,
                            acc_val=acc_val, acc_tr=acc_tr,
                            epochs_run=epochs_run, acc_val_history=acc_hist,
                            epoch_times_s=epoch_times, wall_to_acc_s=wall_to_acc)

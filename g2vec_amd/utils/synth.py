@@ -97,7 +97,7 @@ def module_activity(n_modules: int, shared_frac: float = 0.2) -> np.ndarray:
 
 def synth_expression(genes: Sequence[str], sample_labels: Sequence[int],
                      module: np.ndarray, seed: int,
-                     shared_frac: float = 0.2) -> np.ndarray:
+                     shared_frac: float = 0.0) -> np.ndarray:
     """f32 [S, G] expression with class-dependent module correlation."""
     rng = np.random.default_rng(seed + 777)
     S, G = len(sample_labels), len(genes)
@@ -170,7 +170,7 @@ def synth_network(n_genes: int, n_edges: int, n_modules: int, seed: int,
 
 def synth_dataset(n_genes: int, n_edges: int, n_samples: int,
                   n_modules: int = 24, seed: int = 0, dead_frac: float = 0.5,
-                  shared_frac: float = 0.1) -> Dict:
+                  shared_frac: float = 0.0) -> Dict:
     """Fully synthetic in-memory dataset (scale configs). Returns dict with
     'expr' f32 [S,G], 'labels' i64 [S], 'genes', 'edge_idx' i64 [E,2],
     'module' [G]."""
@@ -218,7 +218,7 @@ def make_ex_style_files(outdir: str, n_genes: int = 7523, n_extra: int = 400,
                         n_edges: int = 298799, n_samples: int = 135,
                         n_poor: int = 58, n_modules: int = 16,
                         seed: int = 0, dead_frac: float = 0.5,
-                        shared_frac: float = 0.1) -> Dict[str, str]:
+                        shared_frac: float = 0.0) -> Dict[str, str]:
     """Emit an ex_*-shaped file triple (expression / clinical / network).
 
     The network carries n_genes + n_extra genes; the expression matrix

@@ -1,0 +1,83 @@
+"""L-groups (incl. the compat bug flag) and biomarker scoring
+(G2Vec.py:85-200)."""
+import numpy as np
+
+from g2vec_amd.cluster import find_lgroups
+from g2vec_amd.scoring import (select_biomarkers, t_scores, t_statistic,
+                               transform_minmax)
+
+
+def _embeddings():
+    rng = np.random.default_rng(0)
+    a = rng.normal(0, 0.05, size=(20, 8)) + np.array([3.] + [0.] * 7)
+    b = rng.normal(0, 0.05, size=(30, 8)) + np.array([0.] * 7 + [3.])
+    c = rng.normal(0, 0.05, size=(100, 8))         # largest cluster -> "other"
+    return np.concatenate([a, b, c]).astype(np.float32)
+
+
+def test_find_lgroups_intended_semantics():
+    emb = _embeddings()
+    freq = np.full(150, 2, dtype=np.int64)
+    freq[:20] = 0      # cluster A genes appear more in good paths
+    freq[20:50] = 1    # cluster B genes more in poor paths
+    lg = find_lgroups(emb, freq, compat_lgroup_bug=False)
+    assert (lg[:20] == 0).all()      # good
+    assert (lg[20:50] == 1).all()    # poor
+    assert (lg[50:] == 2).all()      # other (largest)
+
+
+def test_find_lgroups_bug_mode_ignores_freq():
+    emb = _embeddings()
+    freq = np.full(150, 2, dtype=np.int64)
+    freq[:20] = 0
+    freq[20:50] = 1
+    lg_bug = find_lgroups(emb, freq, compat_lgroup_bug=True)
+    freq_swapped = np.full(150, 2, dtype=np.int64)
+    freq_swapped[:20] = 1
+    freq_swapped[20:50] = 0
+    lg_bug2 = find_lgroups(emb, freq_swapped, compat_lgroup_bug=True)
+    # shipped behaviour never reads freq: same output either way
+    assert (lg_bug == lg_bug2).all()
+    assert set(lg_bug.tolist()) == {0, 1, 2}
+
+
+def test_t_statistic_vs_scipy():
+    from scipy import stats
+    rng = np.random.default_rng(1)
+    x = rng.normal(0.5, 1.0, 40)
+    y = rng.normal(0.0, 1.2, 30)
+    t_ref = stats.ttest_ind(x, y, equal_var=True).statistic
+    assert abs(t_statistic(x, y) - t_ref) < 1e-10
+    assert t_statistic(np.ones(5), np.ones(7)) == 0.0
+
+
+def test_minmax():
+    x = np.array([1.0, 3.0, 2.0])
+    out = transform_minmax(x)
+    assert np.allclose(out, [0.0, 1.0, 0.5])
+    assert (transform_minmax(np.ones(4)) == 0).all()
+
+
+def test_select_biomarkers():
+    rng = np.random.default_rng(2)
+    G, S = 30, 20
+    emb = rng.normal(size=(G, 8)).astype(np.float32)
+    lg = np.array([0] * 10 + [1] * 10 + [2] * 10)
+    expr = rng.normal(size=(S, G)).astype(np.float32)
+    labels = np.array([0] * 10 + [1] * 10)
+    genes = [f"G{i:02d}" for i in range(G)]
+    out = select_biomarkers(emb, lg, expr, labels, genes, num_biomarker=5)
+    assert len(out) == 10                     # 5 per L-group, union
+    assert out == sorted(out)
+    assert all(g in genes[:20] for g in out)  # only good/poor L-group genes
+
+    out_all = select_biomarkers(emb, lg, expr, labels, genes, num_biomarker=50)
+    assert len(out_all) == 20                 # capped by group sizes
+
+
+def test_tscores_shape():
+    rng = np.random.default_rng(3)
+    expr = rng.normal(size=(25, 6)).astype(np.float32)
+    labels = np.array([0] * 12 + [1] * 13)
+    ts = t_scores(expr, labels)
+    assert ts.shape == (6,) and (ts >= 0).all()

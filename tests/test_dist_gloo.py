@@ -1,0 +1,85 @@
+"""Multi-process DP coverage on CPU (gloo, world_size=2): the C1-C5
+collective path must produce the same training result as single-process
+(SURVEY §4.4)."""
+import os
+import socket
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from g2vec_amd.config import G2VecConfig
+from g2vec_amd.models.cbow import CbowTrainer
+from g2vec_amd.parallel.dist import DistContext
+from g2vec_amd.paths import PathSet
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _pathset(G=40, P=160, seed=11):
+    rng = np.random.default_rng(seed)
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 10))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    return PathSet(torch.tensor(genes, dtype=torch.int32),
+                   torch.tensor(offs, dtype=torch.int32),
+                   torch.tensor(labels), G)
+
+
+def _worker(rank, world, port, out):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        ps = _pathset()
+        cfg = G2VecConfig(hidden=64, epochs=6, early_stop=False, seed=4,
+                          device="cpu", dtype="fp32")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                         log=lambda *a, **k: None)
+        res = tr.train(ps)
+
+        # varlen allgather smoke (C5)
+        t = torch.arange((rank + 1) * 3, dtype=torch.float32).reshape(-1, 1)
+        parts = ctx.allgather_varlen(t)
+        assert [p.shape[0] for p in parts] == [3, 6]
+        assert torch.allclose(parts[1][:, 0], torch.arange(6, dtype=torch.float32))
+
+        if rank == 0:
+            out.put((res.W_ih.numpy(), res.acc_val_history))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_matches_single_process():
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    out = ctxm.Queue()
+    procs = [ctxm.Process(target=_worker, args=(r, 2, port, out))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    W_dp, hist_dp = out.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    ps = _pathset()
+    cfg = G2VecConfig(hidden=64, epochs=6, early_stop=False, seed=4,
+                      device="cpu", dtype="fp32")
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **k: None)
+    res = tr.train(ps)
+    assert hist_dp == pytest.approx(res.acc_val_history, abs=1e-6)
+    assert np.allclose(W_dp, res.W_ih.numpy(), atol=1e-5)

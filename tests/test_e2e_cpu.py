@@ -1,0 +1,83 @@
+"""End-to-end pipeline on tiny synthetic ex_*-style files (integration tier,
+SURVEY §4.3)."""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from g2vec_amd.config import G2VecConfig
+from g2vec_amd.io import load_expression
+from g2vec_amd.pipeline import run
+
+
+def _cfg(tiny_files, tmp_path, **kw):
+    base = dict(expression_file=tiny_files["expression"],
+                clinical_file=tiny_files["clinical"],
+                network_file=tiny_files["network"],
+                result_name=str(tmp_path / "out"),
+                len_path=15, num_repetition=3, epochs=25, device="cpu",
+                seed=0)
+    base.update(kw)
+    return G2VecConfig(**base)
+
+
+def test_pipeline_end_to_end(tiny_files, tmp_path):
+    res = run(_cfg(tiny_files, tmp_path))
+    assert res["n_samples"] == 80
+    assert res["n_genes"] == 300
+    assert res["n_paths"] > 50
+    assert 0 < res["n_genes_in_paths"] <= 300
+    assert 0.4 <= res["acc_val"] <= 1.0
+
+    # output files exist and parse in the documented formats
+    bio = (tmp_path / "out_biomarkers.txt").read_text().splitlines()
+    assert bio[0] == "GeneSymbol" and len(bio) > 1
+    assert bio[1:] == sorted(bio[1:])
+
+    lg = (tmp_path / "out_lgroups.txt").read_text().splitlines()
+    assert lg[0] == "GeneSymbol\tLgroup(0:good,1:poor,2:other)"
+    assert len(lg) == 1 + res["n_genes"]
+    vals = {line.split("\t")[1] for line in lg[1:]}
+    assert vals <= {"0", "1", "2"}
+
+    vec = (tmp_path / "out_vectors.txt").read_text().splitlines()
+    assert vec[0].startswith("GeneSymbol\tV0\t")
+    assert len(vec) == 1 + res["n_genes"]
+    assert len(vec[1].split("\t")) == 1 + 128
+
+
+def test_pipeline_compat_bug_flag(tiny_files, tmp_path):
+    res = run(_cfg(tiny_files, tmp_path, compat_lgroup_bug=True))
+    assert set(np.unique(res["lgroups"])) <= {0, 1, 2}
+
+
+def test_save_load_paths(tiny_files, tmp_path):
+    cache = str(tmp_path / "paths.pt")
+    r1 = run(_cfg(tiny_files, tmp_path, save_paths=cache))
+    assert os.path.exists(cache)
+    r2 = run(_cfg(tiny_files, tmp_path, load_paths=cache))
+    assert r2["n_paths"] == r1["n_paths"]
+    assert r2["acc_val"] == pytest.approx(r1["acc_val"], abs=1e-6)
+
+
+def test_cli_surface(tiny_files, tmp_path):
+    from g2vec_amd.cli import args_to_config, build_parser
+    argv = [tiny_files["expression"], tiny_files["clinical"],
+            tiny_files["network"], str(tmp_path / "o"),
+            "-p", "20", "-r", "2", "-s", "64", "-e", "10", "-l", "0.01",
+            "-n", "7", "--device", "cpu", "--compat-lgroup-bug"]
+    cfg = args_to_config(build_parser().parse_args(argv))
+    assert cfg.len_path == 20 and cfg.num_repetition == 2
+    assert cfg.hidden == 64 and cfg.epochs == 10
+    assert cfg.lr == 0.01 and cfg.num_biomarker == 7
+    assert cfg.compat_lgroup_bug
+
+
+def test_jsonl_metrics(tiny_files, tmp_path):
+    log = str(tmp_path / "m.jsonl")
+    run(_cfg(tiny_files, tmp_path, log_jsonl=log))
+    import json
+    events = [json.loads(l) for l in open(log)]
+    names = {e["event"] for e in events}
+    assert {"counts", "paths", "train", "phase"} <= names

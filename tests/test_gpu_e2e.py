@@ -1,0 +1,76 @@
+"""Full-pipeline GPU integration at the ex_* scale (needs an MI355X)."""
+import pytest
+import torch
+
+from g2vec_amd.config import G2VecConfig
+from g2vec_amd.pipeline import run
+from g2vec_amd.utils.synth import make_ex_style_files
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.timeout(900)
+def test_ex_scale_pipeline_on_gpu(tmp_path):
+    files = make_ex_style_files(str(tmp_path), n_genes=7523, n_extra=2381,
+                                n_edges=298799, n_samples=135, n_poor=58,
+                                n_modules=16, seed=0)
+    cfg = G2VecConfig(expression_file=files["expression"],
+                      clinical_file=files["clinical"],
+                      network_file=files["network"],
+                      result_name=str(tmp_path / "out"),
+                      len_path=80, num_repetition=10, epochs=500,
+                      device="cuda", seed=0)
+    res = run(cfg)
+    # README invariants that are deterministic + statistical bands
+    assert res["n_samples"] == 135
+    assert res["n_genes"] == 7523
+    assert 30000 < res["n_paths"] < 80000
+    assert 3000 < res["n_genes_in_paths"] < 4600
+    assert res["acc_val"] >= 0.88          # the BASELINE.md headline bar
+    # the hot phases must have run on the GPU in sane time
+    t = res["timers"]
+    assert t["walks_g0"] + t["walks_g1"] < 60.0
+    assert t["train"] < 120.0
+
+
+@pytest.mark.timeout(600)
+def test_general_trainer_path_on_gpu(tmp_path):
+    files = make_ex_style_files(str(tmp_path), n_genes=1000, n_extra=100,
+                                n_edges=30000, n_samples=100, n_poor=43,
+                                n_modules=8, seed=1)
+    for dtype in ("fp32", "bf16"):
+        cfg = G2VecConfig(expression_file=files["expression"],
+                          clinical_file=files["clinical"],
+                          network_file=files["network"],
+                          result_name=str(tmp_path / f"out_{dtype}"),
+                          len_path=40, num_repetition=5, epochs=20,
+                          device="cuda", seed=0, trainer_path="general",
+                          dtype=dtype)
+        res = run(cfg)
+        assert res["acc_val"] > 0.5
+
+
+@pytest.mark.timeout(600)
+def test_fast_equals_general_on_gpu(tmp_path):
+    import numpy as np
+    from g2vec_amd.models.cbow import CbowTrainer
+    from g2vec_amd.paths import PathSet
+    rng = np.random.default_rng(3)
+    G, P = 200, 400
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 15))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda", 0)
+    ps = PathSet(torch.tensor(genes, dtype=torch.int32, device=dev),
+                 torch.tensor(offs, dtype=torch.int32, device=dev),
+                 torch.tensor(labels, device=dev), G)
+    hists = {}
+    for path in ("fast", "general"):
+        cfg = G2VecConfig(hidden=128, epochs=6, early_stop=False, seed=5,
+                          device="cuda", trainer_path=path, dtype="fp32")
+        res = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None).train(ps)
+        hists[path] = res.acc_val_history
+    assert hists["fast"] == pytest.approx(hists["general"], abs=1e-5)

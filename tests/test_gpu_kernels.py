@@ -1,0 +1,201 @@
+"""HIP kernel parity vs the fp32 CPU oracles (kernel tier).
+All tests need an MI355X (marked gpu)."""
+import numpy as np
+import pytest
+import torch
+
+from g2vec_amd import ops
+from g2vec_amd.ops import cpu_ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _csr(edges, weights, G):
+    e = np.asarray(edges)
+    order = np.lexsort((e[:, 1], e[:, 0]))
+    e = e[order]
+    w = np.asarray(weights, dtype=np.float32)[order]
+    counts = np.bincount(e[:, 0], minlength=G)
+    rp = np.zeros(G + 1, dtype=np.int32)
+    rp[1:] = np.cumsum(counts)
+    return (torch.from_numpy(rp), torch.from_numpy(e[:, 1].astype(np.int32)),
+            torch.from_numpy(w))
+
+
+def test_native_required_on_gpu(monkeypatch):
+    """The GPU path must fail loudly without the extension, never fall back."""
+    monkeypatch.setattr(ops, "_NATIVE", None)
+    with pytest.raises(RuntimeError, match="HIP extension"):
+        ops.native()
+
+
+def test_walks_bitwise_vs_cpu_oracle_uniform_weights():
+    """With exactly-representable uniform weights the float sums are exact,
+    so GPU and CPU walks follow identical RNG decisions bitwise."""
+    rng = np.random.default_rng(0)
+    G = 64
+    edges = np.unique(rng.integers(0, G, size=(600, 2)), axis=0)
+    edges = edges[edges[:, 0] != edges[:, 1]]
+    rp, ci, w = _csr(edges, np.ones(len(edges)), G)
+    src = torch.arange(G, dtype=torch.int32)
+    cn, cl, ch = cpu_ref.random_walks(rp, ci, w, src, 4, 16, seed=123)
+    gn, gl, gh = ops.random_walks(rp.to(DEV), ci.to(DEV), w.to(DEV),
+                                  src.to(DEV), 4, 16, seed=123)
+    assert torch.equal(gl.cpu(), cl)
+    assert torch.equal(gn.cpu(), cn)
+    assert torch.equal(gh.cpu(), ch)
+
+
+def test_walks_structural_random_weights():
+    rng = np.random.default_rng(1)
+    G = 200
+    edges = np.unique(rng.integers(0, G, size=(4000, 2)), axis=0)
+    edges = edges[edges[:, 0] != edges[:, 1]]
+    rp, ci, w = _csr(edges, rng.uniform(0.5, 1.0, len(edges)), G)
+    src = torch.arange(G, dtype=torch.int32)
+    nodes, lengths, hashes = ops.random_walks(rp.to(DEV), ci.to(DEV), w.to(DEV),
+                                              src.to(DEV), 5, 40, seed=7)
+    nodes, lengths = nodes.cpu().numpy(), lengths.cpu().numpy()
+    adj = {(int(a), int(b)) for a, b in edges}
+    for i in range(0, nodes.shape[0], 37):
+        L = lengths[i]
+        path = nodes[i, :L]
+        assert nodes[i, 0] == i % G
+        assert len(set(path.tolist())) == L
+        for k in range(L - 1):
+            assert (int(path[k]), int(path[k + 1])) in adj
+    # hash parity with the CPU hash function
+    for i in range(0, nodes.shape[0], 101):
+        L = lengths[i]
+        assert int(hashes[i]) == int(cpu_ref.path_hash(nodes[i, :L].tolist()))
+
+
+def test_walks_sampling_distribution_gpu():
+    edges = np.array([[0, 1], [0, 2], [0, 3]])
+    rp, ci, w = _csr(edges, [0.6, 0.3, 0.1], 4)
+    n = 20000
+    nodes, _, _ = ops.random_walks(rp.to(DEV), ci.to(DEV), w.to(DEV),
+                                   torch.zeros(1, dtype=torch.int32, device=DEV),
+                                   n, 2, seed=3)
+    first = nodes[:, 1].cpu().numpy()
+    freq = np.bincount(first, minlength=4)[1:4] / n
+    assert np.allclose(freq, [0.6, 0.3, 0.1], atol=0.02)
+
+
+def _pathset_tensors(G=300, P=500, seed=2):
+    rng = np.random.default_rng(seed)
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 30))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    return (torch.tensor(genes, dtype=torch.int32),
+            torch.tensor(offs, dtype=torch.int32),
+            torch.tensor(labels))
+
+
+def test_cbow_fwd_scalar_matches_oracle():
+    genes, offs, labels = _pathset_tensors()
+    s = torch.randn(300)
+    lc, cc, dc = cpu_ref.cbow_fwd_scalar(s, genes, offs, labels, 1 / 500, True)
+    lg, cg, dg = ops.cbow_fwd_scalar(s.to(DEV), genes.to(DEV), offs.to(DEV),
+                                     labels.to(DEV), 1 / 500, True)
+    assert torch.allclose(lg.cpu(), lc, atol=1e-5)
+    assert torch.equal(cg.cpu(), cc)
+    assert torch.allclose(dg.cpu(), dc, atol=1e-7)
+
+
+def test_scatter_do_det_matches_oracle():
+    genes, offs, labels = _pathset_tensors(seed=3)
+    dO = torch.randn(500)
+    c_ref = cpu_ref.scatter_dO(genes, offs, dO, 300)
+    plan = ops.build_scatter_plan(genes.to(DEV), offs.to(DEV), 300)
+    c_gpu = ops.scatter_dO(genes.to(DEV), offs.to(DEV), dO.to(DEV), 300, plan)
+    assert torch.allclose(c_gpu.cpu(), c_ref, atol=1e-5)
+    # determinism: two runs bitwise equal
+    c2 = ops.scatter_dO(genes.to(DEV), offs.to(DEV), dO.to(DEV), 300, plan)
+    assert torch.equal(c_gpu, c2)
+
+
+def test_adam_kernels_match_oracle():
+    torch.manual_seed(0)
+    G, h = 128, 128
+    c = torch.randn(G)
+    who = torch.randn(h)
+    W = torch.randn(G, h)
+    m = torch.rand(G, h) * 0.1
+    v = torch.rand(G, h) * 0.1
+    Wg, mg, vg = (x.clone().to(DEV) for x in (W, m, v))
+    cpu_ref.adam_rank1(W, m, v, c, who, 5, 0.005, 0.9, 0.999, 1e-8)
+    ops.adam_rank1(Wg, mg, vg, c.to(DEV), who.to(DEV), 5, 0.005, 0.9, 0.999, 1e-8)
+    assert torch.allclose(Wg.cpu(), W, atol=1e-6)
+    assert torch.allclose(mg.cpu(), m, atol=1e-7)
+    assert torch.allclose(vg.cpu(), v, atol=1e-7)
+
+    grad = torch.randn(G, h)
+    W2, m2, v2 = torch.randn(G, h), torch.zeros(G, h), torch.zeros(G, h)
+    W2g, m2g, v2g = (x.clone().to(DEV) for x in (W2, m2, v2))
+    cpu_ref.adam_dense(W2, m2, v2, grad, 1, 0.005, 0.9, 0.999, 1e-8)
+    ops.adam_dense(W2g, m2g, v2g, grad.to(DEV), 1, 0.005, 0.9, 0.999, 1e-8)
+    assert torch.allclose(W2g.cpu(), W2, atol=1e-6)
+
+
+@pytest.mark.parametrize("hidden", [64, 128, 256])
+@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+def test_cbow_fwd_general_matches_oracle(hidden, dtype):
+    genes, offs, labels = _pathset_tensors(seed=4)
+    torch.manual_seed(1)
+    W = torch.randn(300, hidden)
+    who = torch.randn(hidden)
+    Wd = W.to(DEV).bfloat16().contiguous() if dtype == "bf16" else W.to(DEV)
+    Wc = Wd.cpu().float()   # oracle sees the same (possibly rounded) weights
+    lc, cc, dc, Hc = cpu_ref.cbow_fwd(Wc, who, genes, offs, labels, 1 / 500, True)
+    lg, cg, dg, Hg = ops.cbow_fwd(Wd, who.to(DEV), genes.to(DEV), offs.to(DEV),
+                                  labels.to(DEV), 1 / 500, True)
+    atol = 1e-4 if dtype == "fp32" else 5e-3
+    assert torch.allclose(Hg.cpu(), Hc, atol=atol)
+    assert torch.allclose(lg.cpu(), lc, atol=atol)
+    assert (cg.cpu() == cc).float().mean() > 0.99   # ties can flip at bf16
+    assert torch.allclose(dg.cpu(), dc, atol=atol)
+
+
+def test_cbow_bwd_rows_matches_oracle():
+    genes, offs, labels = _pathset_tensors(seed=5)
+    who = torch.randn(128)
+    dO = torch.randn(500) * 0.01
+    ref = cpu_ref.cbow_bwd_rows(who, genes, offs, dO, 300)
+    got = ops.cbow_bwd_rows(who.to(DEV), genes.to(DEV), offs.to(DEV),
+                            dO.to(DEV), 300)
+    assert torch.allclose(got.cpu(), ref, atol=1e-5)
+
+
+def test_pcc_edges_matches_corrcoef():
+    rng = np.random.default_rng(6)
+    G, S = 50, 77
+    X = rng.standard_normal((S, G)).astype(np.float32)
+    expr = torch.from_numpy(X)
+    from g2vec_amd.graph import zscore_group
+    zt = zscore_group(expr, torch.zeros(S, dtype=torch.int64), 0)
+    edges = torch.tensor([[i, j] for i in range(10) for j in range(10) if i != j],
+                         dtype=torch.int32)
+    w = ops.pcc_edges(zt.to(DEV), edges.to(DEV), S).cpu().numpy()
+    R = np.corrcoef(X.T)
+    ref = np.array([abs(R[i, j]) for i, j in edges.numpy()])
+    assert np.allclose(w, ref, atol=1e-4)
+
+
+def test_corr_gemm_matches_torch_mm():
+    rng = np.random.default_rng(7)
+    for G, S in ((100, 77), (257, 135)):
+        zt = torch.from_numpy(rng.standard_normal((G, S)).astype(np.float32))
+        C_ref = (zt @ zt.t()) / S
+        C = ops.corr_gemm(zt.to(DEV), S).cpu()
+        assert torch.allclose(C, C_ref, atol=1e-4), (G, S)
+
+
+def test_bf16_copy():
+    x = torch.randn(1000, device=DEV)
+    assert torch.equal(ops.native().bf16_copy(x), x.bfloat16())

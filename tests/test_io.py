@@ -1,0 +1,63 @@
+"""Parser/writer round-trips against the documented formats."""
+import numpy as np
+import pytest
+
+from g2vec_amd.io import (load_clinical, load_expression, load_network,
+                          write_biomarkers, write_lgroups, write_vectors)
+
+
+def _write(p, text):
+    p.write_text(text)
+    return str(p)
+
+
+def test_expression_roundtrip(tmp_path):
+    path = _write(tmp_path / "e.tsv",
+                  "PATIENT\tS1\tS2\nA1BG\t1.5\t-2.25\nB2M\t0.0\t4.5\n")
+    d = load_expression(path)
+    assert list(d["sample"]) == ["S1", "S2"]
+    assert list(d["gene"]) == ["A1BG", "B2M"]
+    # transposed to samples x genes (G2Vec.py:498)
+    assert d["expr"].shape == (2, 2)
+    assert d["expr"][0, 0] == 1.5 and d["expr"][1, 1] == 4.5
+
+
+def test_expression_python_fallback_matches_native(tmp_path):
+    path = _write(tmp_path / "e.tsv",
+                  "PATIENT\tS1\tS2\tS3\nG1\t0.25\t-1\t3\nG2\t2\t0\t-0.5\n")
+    d_native = load_expression(path, use_native=True)
+    d_py = load_expression(path, use_native=False)
+    assert np.allclose(d_native["expr"], d_py["expr"])
+    assert list(d_native["gene"]) == list(d_py["gene"])
+    assert list(d_native["sample"]) == list(d_py["sample"])
+
+
+def test_clinical(tmp_path):
+    path = _write(tmp_path / "c.tsv",
+                  "PATIENT_BARCODE\tLABEL\nS1\t0\nS2\t1\nS3\t0\n")
+    c = load_clinical(path)
+    assert c == {"S1": 0, "S2": 1, "S3": 0}
+
+
+def test_network_directed(tmp_path):
+    path = _write(tmp_path / "n.tsv", "src\tdest\nA\tB\nB\tA\nA\tC\n")
+    n = load_network(path)
+    assert n["edge"] == [("A", "B"), ("B", "A"), ("A", "C")]
+    assert n["gene"] == {"A", "B", "C"}
+
+
+def test_writers_golden(tmp_path):
+    base = str(tmp_path / "res")
+    write_biomarkers(base, ["ADH1C", "AKAP13"])
+    assert (tmp_path / "res_biomarkers.txt").read_text() == \
+        "GeneSymbol\nADH1C\nAKAP13\n"
+
+    write_lgroups(base, [2, 0, 1], ["A1CF", "A2M", "AAK1"])
+    assert (tmp_path / "res_lgroups.txt").read_text() == (
+        "GeneSymbol\tLgroup(0:good,1:poor,2:other)\n"
+        "A1CF\t2\nA2M\t0\nAAK1\t1\n")
+
+    mat = np.array([[0.1234567, -1.0]], dtype=np.float32)
+    write_vectors(base, mat, ["A1CF"])
+    assert (tmp_path / "res_vectors.txt").read_text() == (
+        "GeneSymbol\tV0\tV1\nA1CF\t0.123457\t-1.000000\n")

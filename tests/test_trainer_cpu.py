@@ -1,0 +1,150 @@
+"""CBOW trainer numerics: the collapsed rank-1 algebra vs a dense autograd
+oracle, TF1-Adam formula, fast==general path equivalence, early-stop
+keep-last-good semantics (G2Vec.py:217-286)."""
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from g2vec_amd import ops
+from g2vec_amd.config import G2VecConfig
+from g2vec_amd.models.cbow import CbowTrainer, _trunc_normal
+from g2vec_amd.ops import cpu_ref
+from g2vec_amd.paths import PathSet
+
+
+def _random_pathset(G=50, P=200, seed=0):
+    rng = np.random.default_rng(seed)
+    genes, offs, labels = [], [0], []
+    for p in range(P):
+        L = int(rng.integers(1, 12))
+        gs = rng.choice(G, size=L, replace=False)
+        genes += gs.tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    return PathSet(torch.tensor(genes, dtype=torch.int32),
+                   torch.tensor(offs, dtype=torch.int32),
+                   torch.tensor(labels), G)
+
+
+def _dense_X(ps):
+    X = torch.zeros(ps.n_paths, ps.n_genes)
+    offs = ps.offsets.long()
+    for p in range(ps.n_paths):
+        X[p, ps.genes[offs[p]:offs[p + 1]].long()] = 1.0
+    return X
+
+
+def test_rank1_gradients_match_autograd():
+    """dW_ih = (X^T dO) (x) who and dW_ho = W^T (X^T dO) — the collapsed
+    backward must equal autograd on the dense multi-hot model."""
+    ps = _random_pathset()
+    G, h = ps.n_genes, 64
+    torch.manual_seed(0)
+    W = torch.randn(G, h, requires_grad=True)
+    who = torch.randn(h, requires_grad=True)
+    X = _dense_X(ps)
+    y = ps.labels
+    o = (X @ W) @ who
+    loss = torch.nn.functional.binary_cross_entropy_with_logits(o, y)
+    loss.backward()
+
+    inv_b = 1.0 / ps.n_paths
+    lossv, correct, dO = cpu_ref.cbow_fwd_scalar(
+        torch.mv(W.detach(), who.detach()), ps.genes, ps.offsets, y, inv_b, True)
+    assert abs(float(lossv.mean()) - float(loss)) < 1e-5
+    c = cpu_ref.scatter_dO(ps.genes, ps.offsets, dO, G)
+    dW_fast = torch.outer(c, who.detach())
+    dwho_fast = torch.mv(W.detach().t(), c)
+    assert torch.allclose(dW_fast, W.grad, atol=1e-6)
+    assert torch.allclose(dwho_fast, who.grad, atol=1e-5)
+
+    # general path produces the same gradients
+    l2, c2, dO2, H = cpu_ref.cbow_fwd(W.detach(), who.detach(), ps.genes,
+                                      ps.offsets, y, inv_b, True)
+    dW_gen = cpu_ref.cbow_bwd_rows(who.detach(), ps.genes, ps.offsets, dO2, G)
+    dwho_gen = torch.mv(H.t(), dO2)
+    assert torch.allclose(dW_gen, W.grad, atol=1e-6)
+    assert torch.allclose(dwho_gen, who.grad, atol=1e-5)
+
+
+def test_tf1_adam_formula():
+    """theta -= lr*sqrt(1-b2^t)/(1-b1^t) * m/(sqrt(v)+eps) (TF1 AdamOptimizer)."""
+    torch.manual_seed(1)
+    W = torch.randn(8, 4)
+    W0 = W.clone()
+    m = torch.zeros_like(W)
+    v = torch.zeros_like(W)
+    g = torch.randn(8, 4)
+    lr, b1, b2, eps = 0.005, 0.9, 0.999, 1e-8
+    cpu_ref.adam_dense(W, m, v, g, 1, lr, b1, b2, eps)
+    lr_t = lr * math.sqrt(1 - b2) / (1 - b1)
+    m_ref = 0.1 * g
+    v_ref = 0.001 * g * g
+    W_ref = W0 - lr_t * m_ref / (v_ref.sqrt() + eps)
+    assert torch.allclose(W, W_ref, atol=1e-7)
+    assert torch.allclose(m, m_ref) and torch.allclose(v, v_ref)
+
+
+def test_adam_rank1_equals_dense():
+    torch.manual_seed(2)
+    G, h = 12, 8
+    c = torch.randn(G)
+    who = torch.randn(h)
+    W1 = torch.randn(G, h)
+    W2 = W1.clone()
+    m1, v1 = torch.zeros(G, h), torch.zeros(G, h)
+    m2, v2 = torch.zeros(G, h), torch.zeros(G, h)
+    cpu_ref.adam_rank1(W1, m1, v1, c, who, 3, 0.01, 0.9, 0.999, 1e-8)
+    cpu_ref.adam_dense(W2, m2, v2, torch.outer(c, who), 3, 0.01, 0.9, 0.999, 1e-8)
+    assert torch.allclose(W1, W2)
+
+
+def test_fast_equals_general_training():
+    ps = _random_pathset(G=40, P=150, seed=3)
+    res = {}
+    for path in ("fast", "general"):
+        cfg = G2VecConfig(hidden=64, epochs=8, early_stop=False, seed=5,
+                          device="cpu", trainer_path=path, dtype="fp32")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                         log=lambda *a, **k: None)
+        res[path] = tr.train(ps)
+    assert res["fast"].acc_val_history == pytest.approx(
+        res["general"].acc_val_history, abs=1e-6)
+    assert torch.allclose(res["fast"].W_ih, res["general"].W_ih, atol=1e-4)
+
+
+def test_early_stop_keeps_previous_epoch_weights():
+    ps = _random_pathset(G=30, P=120, seed=7)
+    cfg = G2VecConfig(hidden=64, epochs=60, early_stop=True, seed=1,
+                      device="cpu")
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **k: None)
+    res = tr.train(ps)
+    if res.stop_epoch >= 0:
+        h = res.acc_val_history
+        # stopped at the first strictly-lower epoch; reported value is the
+        # previous epoch's (G2Vec.py:276-279)
+        assert h[res.stop_epoch + 1] < h[res.stop_epoch]
+        assert res.acc_val == pytest.approx(h[res.stop_epoch])
+
+
+def test_trunc_normal_bounds_and_moments():
+    gen = torch.Generator().manual_seed(0)
+    std = 1.0 / math.sqrt(128)
+    x = _trunc_normal((200000,), std, gen)
+    assert float(x.abs().max()) <= 2 * std + 1e-7
+    assert abs(float(x.mean())) < 1e-3
+    # truncated at 2 sigma: variance ~= 0.774 * sigma^2
+    assert abs(float(x.var()) / (std * std) - 0.774) < 0.02
+
+
+def test_minibatch_runs():
+    ps = _random_pathset(G=30, P=100, seed=9)
+    cfg = G2VecConfig(hidden=64, epochs=3, early_stop=False, seed=1,
+                      device="cpu", batch_size=32)
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **k: None)
+    res = tr.train(ps)
+    assert res.epochs_run == 3 and np.isfinite(res.acc_val)
