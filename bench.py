@@ -48,15 +48,16 @@ def log(*a):
     print(*a, file=sys.stderr, flush=True)
 
 
-def build_dataset(seed: int):
+def build_dataset(seed: int, n_genes: int = 7523, n_edges: int = 298799,
+                  n_extra: int = 2381, n_modules: int = 16):
     """In-memory ex_*-shaped dataset (one per rank; weak scaling)."""
-    n_genes, n_extra = 7523, 2381
     rng = np.random.default_rng(seed)
     n_net = n_genes + n_extra
     module = np.full(n_net, -1, dtype=np.int64)
     live = rng.choice(n_genes, size=int(n_genes * 0.5), replace=False)
-    module[live] = rng.integers(0, 16, size=live.size)
-    _, edge_idx, _ = synth.synth_network(n_net, 298799, 16, seed, module=module)
+    module[live] = rng.integers(0, n_modules, size=live.size)
+    _, edge_idx, _ = synth.synth_network(n_net, n_edges, n_modules, seed,
+                                         module=module)
     # restrict to the common genes (network extras have no expression)
     keep = (edge_idx[:, 0] < n_genes) & (edge_idx[:, 1] < n_genes)
     edge_idx = edge_idx[keep]
@@ -77,6 +78,12 @@ def main() -> int:
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--trainer-path", choices=["fast", "general"],
                     default="fast")
+    ap.add_argument("--n-genes", type=int, default=7523,
+                    help="scale configs: 50000 / 200000 / 1000000")
+    ap.add_argument("--n-edges", type=int, default=298799)
+    ap.add_argument("--n-extra", type=int, default=2381)
+    ap.add_argument("--n-modules", type=int, default=16)
+    ap.add_argument("--acc-target-epochs", type=int, default=60)
     args = ap.parse_args()
 
     ctx = init_dist("auto")
@@ -95,7 +102,9 @@ def main() -> int:
                       device=str(device.type), trainer_path=args.trainer_path)
 
     # ---- dataset + graphs + walks (input pipeline; measured, not the metric)
-    expr, labels, edge_idx, n_genes = build_dataset(args.seed + 1000 * rank)
+    expr, labels, edge_idx, n_genes = build_dataset(
+        args.seed + 1000 * rank, args.n_genes, args.n_edges, args.n_extra,
+        args.n_modules)
     expr_t = torch.from_numpy(expr).to(device)
     labels_t = torch.from_numpy(labels).to(device)
     edge_t = torch.from_numpy(edge_idx).to(device)
@@ -122,7 +131,7 @@ def main() -> int:
     wall_to_acc = None
     acc_val = 0.0
     conv_t0 = time.perf_counter()
-    for ep in range(60):
+    for ep in range(args.acc_target_epochs):
         _acc_tr, acc_val = trainer.run_epoch(st)
         if acc_val >= 0.88:
             if on_gpu:
@@ -167,8 +176,8 @@ def main() -> int:
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_PATHS_PER_SEC, 2),
             "dtype": "fp32",
-            "data": "synthetic (ex_* shape: 7523 genes/135 samples/216k edges; "
-                    "random-init weights)",
+            "data": f"synthetic ({args.n_genes} genes/135 samples/"
+                    f"{args.n_edges} network edges; random-init weights)",
             "config": {
                 "model": "g2vec-cbow",
                 "global_batch": n_tr_global,

@@ -45,11 +45,13 @@ def edge_pcc_weights(zt: torch.Tensor, edge_idx: torch.Tensor,
     G = zt.shape[0]
     n_group = zt.shape[1]
     if mode == "auto":
-        # the dense-output GEMM only pays when the edge set is a sizable
-        # fraction of G^2 AND the G^2 f32 matrix fits comfortably in HBM
+        # small G: the MFMA f32 corr GEMM computes all G^2 correlations in
+        # ~0.1 ms and the dense C fits trivially in 288 GB HBM; large G:
+        # the per-edge dot kernel is the only thing that scales (1M genes
+        # -> a 4 TB dense C)
         dense_bytes = 4 * G * G
-        mode = "gemm" if (zt.is_cuda and dense_bytes < 8 << 30 and
-                          edge_idx.shape[0] * 16 > G * G) else "edge"
+        mode = "gemm" if (zt.is_cuda and dense_bytes <= 2 << 30 and
+                          n_group <= 288) else "edge"
     if mode == "gemm":
         C = ops.corr_gemm(zt, n_group)
         return C[edge_idx[:, 0].long(), edge_idx[:, 1].long()].abs()

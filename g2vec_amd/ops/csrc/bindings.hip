@@ -57,11 +57,13 @@ std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col
   auto hashes = torch::empty({n_walks}, opts_l);
   if (n_walks == 0) return {nodes, lengths, hashes};
   const int wpb = 4;                 // 256 threads = 4 waves
-  const size_t lds = (size_t)wpb * len_path * sizeof(int);
+  int tsize = 256;                   // LDS hash set: >= 2x path length, pow2
+  while (tsize < 2 * (int)len_path) tsize <<= 1;
+  const size_t lds = (size_t)wpb * (len_path + tsize) * sizeof(int);
   hipLaunchKernelGGL(walk_kernel, dim3(grid_for(n_walks, wpb)), dim3(256), lds,
                      cur_stream(), row_ptr.data_ptr<int>(), col_idx.data_ptr<int>(),
                      weights.data_ptr<float>(), sources.data_ptr<int>(),
-                     (int)n_src, n_walks, (int)len_path, (uint64_t)seed,
+                     (int)n_src, n_walks, (int)len_path, tsize, (uint64_t)seed,
                      nodes.data_ptr<int>(), lengths.data_ptr<int>(),
                      (long long*)hashes.data_ptr<int64_t>());
   LAUNCH_CHECK();

@@ -77,22 +77,51 @@ __device__ __forceinline__ uint16_t f32_to_bf16(float f) {
 }
 
 // ================================================================= K10: walks
-// One 64-lane wave per walk. Per step: two lane-parallel passes over the
-// CSR row (masked total, then weighted selection by prefix-scan), visited
-// list in LDS (<= len_path entries, broadcast reads). Matches the reference
-// walk semantics exactly (G2Vec.py:328-346); the RNG stream is the
-// framework's own (counter-based, reproducible, shared with the CPU oracle).
+// One 64-lane wave per walk. Visited-set membership is an O(1) LDS
+// open-addressing hash set (the naive O(path_len) visited-list scan per
+// candidate left the CUs ~12% busy — latency-bound on the scan chain,
+// measured via SQ_BUSY_CYCLES). For deg <= 64 the whole CSR row lives in
+// one register per lane, so sampling is a single pass: masked weight ->
+// wave sum -> wave prefix scan -> ballot pick. Matches the reference walk
+// semantics exactly (G2Vec.py:328-346); the RNG stream is the framework's
+// own (counter-based splitmix64, bit-shared with the CPU oracle).
+#define HSET_EMPTY 0xFFFFFFFFu
+
+__device__ __forceinline__ void hset_insert(uint32_t* tab, uint32_t tmask,
+                                            uint32_t v) {
+  uint32_t h = (v * 2654435761u) & tmask;
+  for (;;) {
+    const uint32_t c = tab[h];
+    if (c == v) return;
+    if (c == HSET_EMPTY) { tab[h] = v; return; }
+    h = (h + 1) & tmask;
+  }
+}
+
+__device__ __forceinline__ bool hset_contains(const uint32_t* tab,
+                                              uint32_t tmask, uint32_t v) {
+  uint32_t h = (v * 2654435761u) & tmask;
+  for (;;) {
+    const uint32_t c = tab[h];
+    if (c == v) return true;
+    if (c == HSET_EMPTY) return false;
+    h = (h + 1) & tmask;
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             const float* __restrict__ wgt, const int* __restrict__ sources,
-            int n_src, long long n_walks, int len_path, uint64_t seed,
-            int* __restrict__ out_nodes, int* __restrict__ out_len,
-            long long* __restrict__ out_hash) {
+            int n_src, long long n_walks, int len_path, int tsize,
+            uint64_t seed, int* __restrict__ out_nodes,
+            int* __restrict__ out_len, long long* __restrict__ out_hash) {
   extern __shared__ int smem[];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wib = threadIdx.x >> 6;            // wave-in-block
   const int wpb = blockDim.x >> 6;             // waves per block
-  int* vis = smem + wib * len_path;
+  int* vis = smem + wib * len_path;            // ordered path (output)
+  uint32_t* tab = (uint32_t*)(smem + wpb * len_path) + wib * tsize;
+  const uint32_t tmask = (uint32_t)tsize - 1;
 
   for (long long walk = (long long)blockIdx.x * wpb + wib; walk < n_walks;
        walk += (long long)gridDim.x * wpb) {
@@ -101,66 +130,90 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
     int cur = sources[walk % n_src];
     int plen = 0;
     uint64_t hash = 0;
+    for (int i = lane; i < tsize; i += WAVE) tab[i] = HSET_EMPTY;
 
     for (int step = 0; step < len_path; ++step) {
-      vis[plen] = cur;                         // same value from every lane
+      if (lane == 0) vis[plen] = cur;
       ++plen;
+      hset_insert(tab, tmask, (uint32_t)cur);  // every lane: same probe, same write
       hash += gene_hash_dev((uint32_t)cur);
       const int s = row_ptr[cur], e = row_ptr[cur + 1];
       const int deg = e - s;
       if (deg <= 0) break;
 
-      // pass 1: total unvisited weight
-      float partial = 0.f;
-      for (int j = lane; j < deg; j += WAVE) {
-        const int cand = col_idx[s + j];
-        float w = wgt[s + j];
-        for (int k = 0; k < plen; ++k)
-          if (vis[k] == cand) { w = 0.f; break; }
-        partial += w;
-      }
-      const float tot = wave_sum(partial);
-      const uint64_t r = sm64_next(state);     // drawn even on dead end (oracle parity)
-      if (!(tot > 0.f)) break;
-      const float target = (float)(u01_from(r) * (double)tot);
-
-      // pass 2: first index whose masked-weight running sum exceeds target
-      int chosen = -1;
-      float base = 0.f;
-      for (int j0 = 0; j0 < deg; j0 += WAVE) {
-        const int j = j0 + lane;
+      if (deg <= WAVE) {
+        // fast path: one candidate per lane, fully register-resident
+        int cand = -1;
         float w = 0.f;
-        if (j < deg) {
-          const int cand = col_idx[s + j];
-          w = wgt[s + j];
-          for (int k = 0; k < plen; ++k)
-            if (vis[k] == cand) { w = 0.f; break; }
+        if (lane < deg) {
+          cand = col_idx[s + lane];
+          w = wgt[s + lane];
+          if (hset_contains(tab, tmask, (uint32_t)cand)) w = 0.f;
         }
+        const float tot = wave_sum(w);
+        const uint64_t r = sm64_next(state);   // drawn even on dead end
+        if (!(tot > 0.f)) break;
+        const float target = (float)(u01_from(r) * (double)tot);
         const float scan = wave_incl_scan(w);
-        const float chunk_tot = __shfl(scan, WAVE - 1);
-        const bool hit = (j < deg) && (w > 0.f) &&
-                         (base + scan > target) && (base + scan - w <= target);
-        const unsigned long long m = __ballot(hit);
-        if (m != 0ULL) { chosen = j0 + (__ffsll((long long)m) - 1); break; }
-        base += chunk_tot;
-      }
-      if (chosen < 0) {
-        // numerical tail: target >= running total by rounding -> last unvisited
-        for (int j0 = ((deg - 1) / WAVE) * WAVE; j0 >= 0 && chosen < 0; j0 -= WAVE) {
+        const bool hit = (w > 0.f) && (scan > target) && (scan - w <= target);
+        const unsigned long long mh = __ballot(hit);
+        int lane_sel;
+        if (mh != 0ULL) {
+          lane_sel = __ffsll((long long)mh) - 1;      // first crossing lane
+        } else {
+          // rounding tail: target >= total running sum -> last unvisited
+          const unsigned long long mp = __ballot(w > 0.f);
+          lane_sel = 63 - __clzll((long long)mp);
+        }
+        cur = __shfl(cand, lane_sel);
+      } else {
+        // chunked path for high-degree nodes
+        float partial = 0.f;
+        for (int j = lane; j < deg; j += WAVE) {
+          const int c2 = col_idx[s + j];
+          float w = wgt[s + j];
+          if (hset_contains(tab, tmask, (uint32_t)c2)) w = 0.f;
+          partial += w;
+        }
+        const float tot = wave_sum(partial);
+        const uint64_t r = sm64_next(state);
+        if (!(tot > 0.f)) break;
+        const float target = (float)(u01_from(r) * (double)tot);
+        int chosen = -1;
+        float base = 0.f;
+        for (int j0 = 0; j0 < deg; j0 += WAVE) {
           const int j = j0 + lane;
           float w = 0.f;
           if (j < deg) {
-            const int cand = col_idx[s + j];
+            const int c2 = col_idx[s + j];
             w = wgt[s + j];
-            for (int k = 0; k < plen; ++k)
-              if (vis[k] == cand) { w = 0.f; break; }
+            if (hset_contains(tab, tmask, (uint32_t)c2)) w = 0.f;
           }
-          const unsigned long long m = __ballot(w > 0.f);
-          if (m != 0ULL) chosen = j0 + (63 - __clzll((long long)m));
+          const float scan = wave_incl_scan(w);
+          const float chunk_tot = __shfl(scan, WAVE - 1);
+          const bool hit = (j < deg) && (w > 0.f) &&
+                           (base + scan > target) && (base + scan - w <= target);
+          const unsigned long long m = __ballot(hit);
+          if (m != 0ULL) { chosen = j0 + (__ffsll((long long)m) - 1); break; }
+          base += chunk_tot;
         }
-        if (chosen < 0) break;                 // cannot happen when tot > 0
+        if (chosen < 0) {
+          for (int j0 = ((deg - 1) / WAVE) * WAVE; j0 >= 0 && chosen < 0;
+               j0 -= WAVE) {
+            const int j = j0 + lane;
+            float w = 0.f;
+            if (j < deg) {
+              const int c2 = col_idx[s + j];
+              w = wgt[s + j];
+              if (hset_contains(tab, tmask, (uint32_t)c2)) w = 0.f;
+            }
+            const unsigned long long m = __ballot(w > 0.f);
+            if (m != 0ULL) chosen = j0 + (63 - __clzll((long long)m));
+          }
+          if (chosen < 0) break;               // cannot happen when tot > 0
+        }
+        cur = col_idx[s + chosen];
       }
-      cur = col_idx[s + chosen];
     }
 
     long long outb = walk * (long long)len_path;
