@@ -7,7 +7,7 @@ CPU tensors -> the torch/numpy oracles in `cpu_ref`.
 """
 from __future__ import annotations
 
-from typing import NamedTuple, Optional, Tuple
+from typing import NamedTuple, Optional
 
 import torch
 

@@ -8,7 +8,7 @@ Reference semantics (all host-side):
 """
 from __future__ import annotations
 
-from typing import Dict, List, Tuple
+from typing import Dict, List
 
 import numpy as np
 
