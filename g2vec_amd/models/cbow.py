@@ -136,11 +136,17 @@ class CbowTrainer:
         bs = cfg.batch_size if cfg.batch_size > 0 else P_loc
         st.batches = [(i, min(i + bs, P_loc)) for i in range(0, max(P_loc, 1), bs)]
         st.t_adam = 0
+        st.s_cache = None
         return st
 
     def run_epoch(self, st) -> tuple:
         """One reference epoch: optimizer step(s) at W_t, then post-update
-        accuracy on both splits (G2Vec.py:262-267). Returns (acc_tr, acc_val)."""
+        accuracy on both splits (G2Vec.py:262-267). Returns (acc_tr, acc_val).
+
+        Fast path: s = W_ih @ W_ho is computed once per weight version —
+        the post-update s used by the accuracy evals is the same one the
+        next epoch's full-batch step needs (full-batch only; minibatching
+        changes W within the epoch and disables the cache)."""
         cfg = self.cfg
         for (lo, hi) in st.batches:
             b_inv = st.inv_b if cfg.batch_size == 0 else 1.0 / (
@@ -150,10 +156,26 @@ class CbowTrainer:
                 self._step_general(st.W, st.W16, st.who, st.mW, st.vW, st.mO,
                                    st.vO, st.tr, lo, hi, b_inv, st.t_adam)
             else:
-                self._step_fast(st.W, st.who, st.mW, st.vW, st.mO, st.vO,
-                                st.tr, st.plan, lo, hi, b_inv, st.t_adam)
-        acc_tr = self._accuracy(st.W, st.W16, st.who, st.tr, self.n_tr_global)
-        acc_val = self._accuracy(st.W, st.W16, st.who, st.vl, self.n_vl_global)
+                self._step_fast(st, lo, hi, b_inv, st.t_adam)
+        if cfg.trainer_path == "general":
+            acc_tr = self._accuracy(st.W, st.W16, st.who, st.tr, self.n_tr_global)
+            acc_val = self._accuracy(st.W, st.W16, st.who, st.vl, self.n_vl_global)
+        else:
+            s = torch.mv(st.W, st.who)          # post-update s
+            if cfg.batch_size == 0:
+                st.s_cache = s                  # reused by the next epoch's step
+            counts = torch.empty(2, dtype=torch.float32, device=self.device)
+            for k, split in enumerate((st.tr, st.vl)):
+                if split.n_paths == 0:
+                    counts[k] = 0.0
+                    continue
+                _l, corr, _d = ops.cbow_fwd_scalar(
+                    s, split.genes, split.offsets, split.labels, 1.0, False)
+                counts[k] = corr.sum()
+            self.ctx.allreduce_(counts)         # C3: one fused metric reduce
+            cc = counts.cpu()
+            acc_tr = float(cc[0]) / max(self.n_tr_global, 1)
+            acc_val = float(cc[1]) / max(self.n_vl_global, 1)
         return acc_tr, acc_val
 
     # ------------------------------------------------------------------ train
@@ -211,18 +233,22 @@ class CbowTrainer:
         o = (ps.offsets[lo:hi + 1] - ps.offsets[lo]).contiguous()
         return g, o, ps.labels[lo:hi]
 
-    def _step_fast(self, W, who, mW, vW, mO, vO, tr, plan, lo, hi, inv_b, t):
+    def _step_fast(self, st, lo, hi, inv_b, t):
+        W, who, tr = st.W, st.who, st.tr
         genes, offsets, labels = self._slice(tr, lo, hi)
-        s = torch.mv(W, who)
+        s = getattr(st, "s_cache", None)
+        if s is None:
+            s = torch.mv(W, who)
+        st.s_cache = None                           # consumed (W changes below)
         _loss, _corr, dO = ops.cbow_fwd_scalar(s, genes, offsets, labels,
                                                inv_b, True)
-        use_plan = plan if (lo == 0 and hi == tr.n_paths) else None
+        use_plan = st.plan if (lo == 0 and hi == tr.n_paths) else None
         c = ops.scatter_dO(genes, offsets, dO, self.G, plan=use_plan)
         self.ctx.allreduce_(c)                      # C1: the whole dW_ih message
         grad_who = torch.mv(W.t(), c)               # dW_ho = W_ih^T c (pre-update W)
-        ops.adam_rank1(W, mW, vW, c, who, t, self.cfg.lr, self.B1, self.B2,
-                       self.EPS)
-        ops.adam_dense(who, mO, vO, grad_who, t, self.cfg.lr, self.B1,
+        ops.adam_rank1(W, st.mW, st.vW, c, who, t, self.cfg.lr, self.B1,
+                       self.B2, self.EPS)
+        ops.adam_dense(who, st.mO, st.vO, grad_who, t, self.cfg.lr, self.B1,
                        self.B2, self.EPS)
 
     def _step_general(self, W, W16, who, mW, vW, mO, vO, tr, lo, hi, inv_b, t):
