@@ -44,6 +44,8 @@ def build_parser() -> argparse.ArgumentParser:
                    help="reproduce the shipped L-group disambiguation bug "
                         "(G2Vec.py:186-194, SURVEY 2.9)")
     p.add_argument("--no-early-stop", action="store_true")
+    p.add_argument("--no-hipgraph", action="store_true",
+                   help="disable hipGraph capture of the training epoch")
     p.add_argument("--save-paths", type=str, default="")
     p.add_argument("--load-paths", type=str, default="")
     p.add_argument("--log-jsonl", type=str, default="")
@@ -61,7 +63,8 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         pcc_mode=a.pcc_mode, trainer_path=a.trainer_path,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, save_paths=a.save_paths,
-        load_paths=a.load_paths, log_jsonl=a.log_jsonl)
+        load_paths=a.load_paths, log_jsonl=a.log_jsonl,
+        use_hipgraph=not a.no_hipgraph)
 
 
 def main(argv=None) -> int:

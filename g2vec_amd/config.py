@@ -39,6 +39,7 @@ class G2VecConfig:
     load_paths: str = ""
     log_jsonl: str = ""             # structured metrics sink
     deterministic_grads: bool = False  # bitwise-reproducible dW_ih reduction (no atomics)
+    use_hipgraph: bool = True       # record the full-batch epoch into a hipGraph
 
     def validate(self) -> None:
         if self.hidden % 64 != 0 or not (64 <= self.hidden <= 1024):

@@ -136,18 +136,80 @@ class CbowTrainer:
         bs = cfg.batch_size if cfg.batch_size > 0 else P_loc
         st.batches = [(i, min(i + bs, P_loc)) for i in range(0, max(P_loc, 1), bs)]
         st.t_adam = 0
-        st.s_cache = None
+        st.epoch_idx = 0
+        # persistent fast-path buffers (stable addresses across hipGraph replays)
+        st.s_buf = torch.mv(W, who) if not use_general else None
+        st.lrt_buf = torch.zeros(1, dtype=torch.float32, device=self.device)
+        st.counts_buf = torch.zeros(2, dtype=torch.float32, device=self.device)
+        st.graph = None
+        st.graph_failed = False
         return st
+
+    def _epoch_body_fast(self, st) -> None:
+        """One full-batch fast-path epoch as a capturable body: optimizer
+        step at W_t, then post-update s + accuracy counts. All inputs and
+        outputs live in persistent buffers (s_buf, lrt_buf, counts_buf) so
+        the body can be recorded once into a hipGraph and replayed."""
+        cfg = self.cfg
+        tr, vl = st.tr, st.vl
+        st.counts_buf.zero_()
+        lrt = st.lrt_buf if self.device.type == "cuda" else None
+        _loss, _corr, dO = ops.cbow_fwd_scalar(
+            st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True)
+        c = ops.scatter_dO(tr.genes, tr.offsets, dO, self.G, plan=st.plan)
+        self.ctx.allreduce_(c)                  # C1: whole dW_ih message
+        grad_who = torch.mv(st.W.t(), c)        # dW_ho = W_ih^T c (pre-update W)
+        ops.adam_rank1(st.W, st.mW, st.vW, c, st.who, st.t_adam, cfg.lr,
+                       self.B1, self.B2, self.EPS, lrt_buf=lrt)
+        ops.adam_dense(st.who, st.mO, st.vO, grad_who, st.t_adam, cfg.lr,
+                       self.B1, self.B2, self.EPS, lrt_buf=lrt)
+        # post-update accuracy (reference order, G2Vec.py:264-267)
+        torch.mv(st.W, st.who, out=st.s_buf)
+        for k, split in enumerate((tr, vl)):
+            if split.n_paths == 0:
+                continue
+            _l, corr, _d = ops.cbow_fwd_scalar(
+                st.s_buf, split.genes, split.offsets, split.labels, 1.0, False)
+            st.counts_buf[k].copy_(corr.sum())
+        self.ctx.allreduce_(st.counts_buf)      # C3: one fused metric reduce
 
     def run_epoch(self, st) -> tuple:
         """One reference epoch: optimizer step(s) at W_t, then post-update
         accuracy on both splits (G2Vec.py:262-267). Returns (acc_tr, acc_val).
 
-        Fast path: s = W_ih @ W_ho is computed once per weight version —
-        the post-update s used by the accuracy evals is the same one the
-        next epoch's full-batch step needs (full-batch only; minibatching
-        changes W within the epoch and disables the cache)."""
+        Fast full-batch path: the whole epoch body runs from persistent
+        buffers; s = W_ih @ W_ho is computed once per weight version, and on
+        GPU the body is recorded into a hipGraph once (epoch >= 2) and
+        replayed thereafter — one graph launch + one 8-byte D2H per epoch."""
         cfg = self.cfg
+        fast_full = cfg.trainer_path != "general" and cfg.batch_size == 0
+        if fast_full:
+            st.t_adam += 1
+            on_gpu = self.device.type == "cuda"
+            if on_gpu:
+                st.lrt_buf.fill_(ops.tf1_lr_t(cfg.lr, self.B1, self.B2,
+                                              st.t_adam))
+            if (on_gpu and cfg.use_hipgraph and st.graph is None and
+                    not st.graph_failed and st.epoch_idx >= 1):
+                try:
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        self._epoch_body_fast(st)
+                    st.graph = g        # capture records without executing
+                except Exception as e:  # noqa: BLE001
+                    st.graph_failed = True
+                    self.log(f"    (hipGraph capture unavailable: {e!r}; "
+                             f"running eager)")
+            if st.graph is not None:
+                st.graph.replay()
+            else:
+                self._epoch_body_fast(st)
+            cc = st.counts_buf.cpu()
+            acc_tr = float(cc[0]) / max(self.n_tr_global, 1)
+            acc_val = float(cc[1]) / max(self.n_vl_global, 1)
+            st.epoch_idx += 1
+            return acc_tr, acc_val
+
         for (lo, hi) in st.batches:
             b_inv = st.inv_b if cfg.batch_size == 0 else 1.0 / (
                 (hi - lo) * self.ctx.world)
@@ -161,21 +223,20 @@ class CbowTrainer:
             acc_tr = self._accuracy(st.W, st.W16, st.who, st.tr, self.n_tr_global)
             acc_val = self._accuracy(st.W, st.W16, st.who, st.vl, self.n_vl_global)
         else:
-            s = torch.mv(st.W, st.who)          # post-update s
-            if cfg.batch_size == 0:
-                st.s_cache = s                  # reused by the next epoch's step
+            torch.mv(st.W, st.who, out=st.s_buf)    # post-update s
             counts = torch.empty(2, dtype=torch.float32, device=self.device)
             for k, split in enumerate((st.tr, st.vl)):
                 if split.n_paths == 0:
                     counts[k] = 0.0
                     continue
                 _l, corr, _d = ops.cbow_fwd_scalar(
-                    s, split.genes, split.offsets, split.labels, 1.0, False)
+                    st.s_buf, split.genes, split.offsets, split.labels, 1.0, False)
                 counts[k] = corr.sum()
             self.ctx.allreduce_(counts)         # C3: one fused metric reduce
             cc = counts.cpu()
             acc_tr = float(cc[0]) / max(self.n_tr_global, 1)
             acc_val = float(cc[1]) / max(self.n_vl_global, 1)
+        st.epoch_idx += 1
         return acc_tr, acc_val
 
     # ------------------------------------------------------------------ train
@@ -234,14 +295,14 @@ class CbowTrainer:
         return g, o, ps.labels[lo:hi]
 
     def _step_fast(self, st, lo, hi, inv_b, t):
+        """Minibatch fast step (full-batch epochs go through
+        _epoch_body_fast instead)."""
         W, who, tr = st.W, st.who, st.tr
         genes, offsets, labels = self._slice(tr, lo, hi)
-        s = getattr(st, "s_cache", None)
-        if s is None:
-            s = torch.mv(W, who)
-        st.s_cache = None                           # consumed (W changes below)
-        _loss, _corr, dO = ops.cbow_fwd_scalar(s, genes, offsets, labels,
-                                               inv_b, True)
+        if lo != 0:                 # s_buf is fresh only for the first batch
+            torch.mv(W, who, out=st.s_buf)
+        _loss, _corr, dO = ops.cbow_fwd_scalar(st.s_buf, genes, offsets,
+                                               labels, inv_b, True)
         use_plan = st.plan if (lo == 0 and hi == tr.n_paths) else None
         c = ops.scatter_dO(genes, offsets, dO, self.G, plan=use_plan)
         self.ctx.allreduce_(c)                      # C1: the whole dW_ih message

@@ -87,20 +87,34 @@ def scatter_dO(genes, offsets, dO, n_genes: int,
     return cpu_ref.scatter_dO(genes, offsets, dO, n_genes)
 
 
+def tf1_lr_t(lr: float, b1: float, b2: float, t: int) -> float:
+    """TF1 AdamOptimizer effective step size (G2Vec.py:245-246 semantics)."""
+    return lr * (1.0 - b2 ** t) ** 0.5 / (1.0 - b1 ** t)
+
+
 def adam_rank1(W, m, v, c, who, t: int, lr: float, b1: float, b2: float,
-               eps: float) -> None:
+               eps: float, lrt_buf: Optional[torch.Tensor] = None) -> None:
+    """lrt_buf: optional persistent f32[1] device buffer holding lr_t —
+    required under hipGraph capture (the per-step value must be read from
+    device memory, not baked into the recorded launch)."""
     if W.is_cuda:
-        native().adam_rank1(W, m, v, c, who, int(t), float(lr), float(b1),
-                            float(b2), float(eps))
+        if lrt_buf is None:
+            lrt_buf = torch.tensor([tf1_lr_t(lr, b1, b2, t)],
+                                   dtype=torch.float32, device=W.device)
+        native().adam_rank1(W, m, v, c, who, lrt_buf, float(b1), float(b2),
+                            float(eps))
         return
     cpu_ref.adam_rank1(W, m, v, c, who, t, lr, b1, b2, eps)
 
 
 def adam_dense(W, m, v, grad, t: int, lr: float, b1: float, b2: float,
-               eps: float) -> None:
+               eps: float, lrt_buf: Optional[torch.Tensor] = None) -> None:
     if W.is_cuda:
-        native().adam_dense(W, m, v, grad, int(t), float(lr), float(b1),
-                            float(b2), float(eps))
+        if lrt_buf is None:
+            lrt_buf = torch.tensor([tf1_lr_t(lr, b1, b2, t)],
+                                   dtype=torch.float32, device=W.device)
+        native().adam_dense(W, m, v, grad, lrt_buf, float(b1), float(b2),
+                            float(eps))
         return
     cpu_ref.adam_dense(W, m, v, grad, t, lr, b1, b2, eps)
 

@@ -113,41 +113,36 @@ torch::Tensor scatter_dO_det(torch::Tensor inst_path, torch::Tensor seg_start,
   return c;
 }
 
-static double tf1_lr_t(double lr, double b1, double b2, int64_t t) {
-  return lr * std::sqrt(1.0 - std::pow(b2, (double)t)) /
-         (1.0 - std::pow(b1, (double)t));
-}
-
 void adam_rank1(torch::Tensor W, torch::Tensor m, torch::Tensor v,
-                torch::Tensor c, torch::Tensor who, int64_t t, double lr,
+                torch::Tensor c, torch::Tensor who, torch::Tensor lrt,
                 double b1, double b2, double eps) {
   CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
   CHECK_CONT(m); CHECK_CONT(v); CHECK_CONT(c); CHECK_CONT(who);
+  CHECK_DEV(lrt); CHECK_F32(lrt);
   const long long G = W.size(0);
   const int h = (int)W.size(1);
   TORCH_CHECK(h % 4 == 0, "hidden must be a multiple of 4");
-  const double lr_t = tf1_lr_t(lr, b1, b2, t);
   const long long n4 = G * h / 4;
   hipLaunchKernelGGL(adam_rank1_kernel, dim3(grid_for(n4, 256)), dim3(256), 0,
                      cur_stream(), W.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), c.data_ptr<float>(),
-                     who.data_ptr<float>(), G, h, (float)lr_t, (float)b1,
-                     (float)b2, (float)eps);
+                     who.data_ptr<float>(), G, h, lrt.data_ptr<float>(),
+                     (float)b1, (float)b2, (float)eps);
   LAUNCH_CHECK();
 }
 
 void adam_dense(torch::Tensor W, torch::Tensor m, torch::Tensor v,
-                torch::Tensor grad, int64_t t, double lr, double b1, double b2,
+                torch::Tensor grad, torch::Tensor lrt, double b1, double b2,
                 double eps) {
   CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
   CHECK_CONT(m); CHECK_CONT(v); CHECK_CONT(grad);
+  CHECK_DEV(lrt); CHECK_F32(lrt);
   TORCH_CHECK(grad.numel() == W.numel(), "grad/W size mismatch");
   const long long n = W.numel();
-  const double lr_t = tf1_lr_t(lr, b1, b2, t);
   hipLaunchKernelGGL(adam_dense_kernel, dim3(grid_for(n, 256)), dim3(256), 0,
                      cur_stream(), W.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), grad.data_ptr<float>(), n,
-                     (float)lr_t, (float)b1, (float)b2, (float)eps);
+                     lrt.data_ptr<float>(), (float)b1, (float)b2, (float)eps);
   LAUNCH_CHECK();
 }
 

@@ -283,7 +283,10 @@ extern "C" __global__ void __launch_bounds__(256)
 adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
                   float* __restrict__ v, const float* __restrict__ c,
                   const float* __restrict__ who, long long G, int h,
-                  float lr_t, float b1, float b2, float eps) {
+                  const float* __restrict__ lr_t_ptr, float b1, float b2,
+                  float eps) {
+  const float lr_t = lr_t_ptr[0];   // device-read so hipGraph replays see
+                                    // the per-step bias-corrected value
   const long long n4 = (long long)G * h / 4;
   f32x4* W4 = (f32x4*)W; f32x4* m4 = (f32x4*)m; f32x4* v4 = (f32x4*)v;
   const f32x4* who4 = (const f32x4*)who;
@@ -309,7 +312,9 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
 extern "C" __global__ void __launch_bounds__(256)
 adam_dense_kernel(float* __restrict__ W, float* __restrict__ m,
                   float* __restrict__ v, const float* __restrict__ grad,
-                  long long n, float lr_t, float b1, float b2, float eps) {
+                  long long n, const float* __restrict__ lr_t_ptr, float b1,
+                  float b2, float eps) {
+  const float lr_t = lr_t_ptr[0];
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long long)gridDim.x * blockDim.x) {
     const float g = grad[i];
