@@ -34,7 +34,7 @@ def build_parser() -> argparse.ArgumentParser:
     # framework flags (absent in reference)
     p.add_argument("--seed", type=int, default=0,
                    help="-1 = unseeded (reference-like nondeterminism)")
-    p.add_argument("--dtype", choices=["fp32", "bf16"], default="bf16")
+    p.add_argument("--dtype", choices=["fp32", "bf16", "fp16"], default="bf16")
     p.add_argument("--device", choices=["auto", "cpu", "cuda"], default="auto")
     p.add_argument("--pcc-mode", choices=["auto", "edge", "gemm"], default="auto")
     p.add_argument("--trainer-path", choices=["fast", "general"], default="fast")

@@ -26,7 +26,7 @@ class G2VecConfig:
 
     # --- framework knobs (absent in reference) ---
     seed: Optional[int] = 0         # None -> nondeterministic like the reference
-    dtype: str = "bf16"             # compute dtype of the W_ih gather: {"fp32","bf16"}
+    dtype: str = "bf16"             # compute dtype of the W_ih gather: {"fp32","bf16","fp16"}
     device: str = "auto"            # "auto" | "cpu" | "cuda"
     pcc_threshold: float = 0.5      # |PCC| cutoff (G2Vec.py:385-390)
     pcc_mode: str = "auto"          # "edge" (per-edge dot) | "gemm" (MFMA corr GEMM) | "auto"
@@ -46,8 +46,8 @@ class G2VecConfig:
             raise ValueError(
                 f"hidden={self.hidden}: MI355X kernels require a multiple of the "
                 f"64-lane wavefront in [64, 1024]")
-        if self.dtype not in ("fp32", "bf16"):
-            raise ValueError(f"dtype must be fp32|bf16, got {self.dtype}")
+        if self.dtype not in ("fp32", "bf16", "fp16"):
+            raise ValueError(f"dtype must be fp32|bf16|fp16, got {self.dtype}")
         if self.len_path < 1 or self.len_path > 512:
             raise ValueError("len_path must be in [1, 512] (LDS visited-list budget)")
         if self.pcc_mode not in ("auto", "edge", "gemm"):

@@ -127,7 +127,9 @@ class CbowTrainer:
         st.mO, st.vO = torch.zeros_like(who), torch.zeros_like(who)
         st.W_keep = W.clone()
         st.tr, st.vl = tr, vl
-        st.W16 = W.bfloat16() if (use_general and cfg.dtype == "bf16") else None
+        st.W16 = None
+        if use_general and cfg.dtype != "fp32":
+            st.W16 = (W.bfloat16() if cfg.dtype == "bf16" else W.half())
         st.plan = None
         if not use_general:
             st.plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)

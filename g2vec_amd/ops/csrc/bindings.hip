@@ -157,7 +157,9 @@ std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
   CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
   CHECK_DEV(labels); CHECK_CONT(labels); CHECK_F32(labels);
   const bool bf16 = W.scalar_type() == at::kBFloat16;
-  TORCH_CHECK(bf16 || W.scalar_type() == at::kFloat, "W must be f32 or bf16");
+  const bool fp16 = W.scalar_type() == at::kHalf;
+  TORCH_CHECK(bf16 || fp16 || W.scalar_type() == at::kFloat,
+              "W must be f32, bf16 or fp16");
   const long long P = labels.numel();
   const int h = (int)W.size(1);
   const int hpl = h / 64;
@@ -180,13 +182,22 @@ std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
                      labels.data_ptr<float>(), P, (float)inv_b, h, Hp,        \
                      loss.data_ptr<float>(), correct.data_ptr<float>(), dOp)
   if (bf16) {
-    const uint16_t* Wp = (const uint16_t*)W.data_ptr<at::BFloat16>();
+    const bf16_bits* Wp = (const bf16_bits*)W.data_ptr<at::BFloat16>();
     switch (hpl) {
-      case 1: FWD_CASE(uint16_t, 1, Wp); break;
-      case 2: FWD_CASE(uint16_t, 2, Wp); break;
-      case 4: FWD_CASE(uint16_t, 4, Wp); break;
-      case 8: FWD_CASE(uint16_t, 8, Wp); break;
-      default: FWD_CASE(uint16_t, 16, Wp); break;
+      case 1: FWD_CASE(bf16_bits, 1, Wp); break;
+      case 2: FWD_CASE(bf16_bits, 2, Wp); break;
+      case 4: FWD_CASE(bf16_bits, 4, Wp); break;
+      case 8: FWD_CASE(bf16_bits, 8, Wp); break;
+      default: FWD_CASE(bf16_bits, 16, Wp); break;
+    }
+  } else if (fp16) {
+    const fp16_bits* Wp = (const fp16_bits*)W.data_ptr<at::Half>();
+    switch (hpl) {
+      case 1: FWD_CASE(fp16_bits, 1, Wp); break;
+      case 2: FWD_CASE(fp16_bits, 2, Wp); break;
+      case 4: FWD_CASE(fp16_bits, 4, Wp); break;
+      case 8: FWD_CASE(fp16_bits, 8, Wp); break;
+      default: FWD_CASE(fp16_bits, 16, Wp); break;
     }
   } else {
     const float* Wp = W.data_ptr<float>();

@@ -22,8 +22,20 @@
 // Reference behaviour being reimplemented (not copied): mathcom/G2Vec
 // G2Vec.py:328-346 (walk), :238-251 (CBOW fwd/loss/acc), :245-246 (Adam),
 // :354-368 (PCC).
+#include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 #include <cstdint>
+
+// Debug build (G2VEC_DEBUG=1 at setup time): device-side bounds asserts in
+// the index-following kernels (SURVEY 5.2 — the debug tier of the race/
+// sanitizer story; host-side ASAN comes from the same flag's -fsanitize
+// on the bindings TU).
+#ifdef G2VEC_DEBUG
+#define G2V_ASSERT(cond) \
+  do { if (!(cond)) __builtin_trap(); } while (0)
+#else
+#define G2V_ASSERT(cond) do { } while (0)
+#endif
 
 #define WAVE 64
 
@@ -66,6 +78,18 @@ __device__ __forceinline__ float bf16_to_f32(uint16_t b) {
   x.u = ((uint32_t)b) << 16;
   return x.f;
 }
+
+// tag types so the gather kernel can distinguish the two 16-bit formats
+struct bf16_bits { uint16_t v; };
+struct fp16_bits { uint16_t v; };
+
+__device__ __forceinline__ float to_f32(const bf16_bits b) {
+  return bf16_to_f32(b.v);
+}
+__device__ __forceinline__ float to_f32(const fp16_bits h) {
+  return __half2float(*(const __half*)&h.v);
+}
+__device__ __forceinline__ float to_f32(const float f) { return f; }
 
 __device__ __forceinline__ uint16_t f32_to_bf16(float f) {
   union { float f; uint32_t u; } x;
@@ -133,6 +157,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
     for (int i = lane; i < tsize; i += WAVE) tab[i] = HSET_EMPTY;
 
     for (int step = 0; step < len_path; ++step) {
+      G2V_ASSERT(plen < len_path && cur >= 0);
       if (lane == 0) vis[plen] = cur;
       ++plen;
       hset_insert(tab, tmask, (uint32_t)cur);  // every lane: same probe, same write
@@ -350,16 +375,12 @@ cbow_fwd_kernel(const WT* __restrict__ W, const float* __restrict__ who,
     for (int k = 0; k < HPL; ++k) acc[k] = 0.f;
     for (int i = lo; i < hi; ++i) {
       const long long g = genes[i];
+      G2V_ASSERT(g >= 0);
       const WT* row = W + g * (long long)h + col0;
       WT tmp[HPL];
       __builtin_memcpy(tmp, row, HPL * sizeof(WT));  // one wide load
 #pragma unroll
-      for (int k = 0; k < HPL; ++k) {
-        if constexpr (sizeof(WT) == 2)
-          acc[k] += bf16_to_f32(*(const uint16_t*)&tmp[k]);
-        else
-          acc[k] += *(const float*)&tmp[k];
-      }
+      for (int k = 0; k < HPL; ++k) acc[k] += to_f32(tmp[k]);
     }
     float op = 0.f;
     float wv[HPL];
@@ -517,11 +538,16 @@ INSTANTIATE_FWD(float, 2)
 INSTANTIATE_FWD(float, 4)
 INSTANTIATE_FWD(float, 8)
 INSTANTIATE_FWD(float, 16)
-INSTANTIATE_FWD(uint16_t, 1)
-INSTANTIATE_FWD(uint16_t, 2)
-INSTANTIATE_FWD(uint16_t, 4)
-INSTANTIATE_FWD(uint16_t, 8)
-INSTANTIATE_FWD(uint16_t, 16)
+INSTANTIATE_FWD(bf16_bits, 1)
+INSTANTIATE_FWD(bf16_bits, 2)
+INSTANTIATE_FWD(bf16_bits, 4)
+INSTANTIATE_FWD(bf16_bits, 8)
+INSTANTIATE_FWD(bf16_bits, 16)
+INSTANTIATE_FWD(fp16_bits, 1)
+INSTANTIATE_FWD(fp16_bits, 2)
+INSTANTIATE_FWD(fp16_bits, 4)
+INSTANTIATE_FWD(fp16_bits, 8)
+INSTANTIATE_FWD(fp16_bits, 16)
 
 #define INSTANTIATE_BWD(HPL)                                                  \
   template __global__ void cbow_bwd_rows_kernel<HPL>(                         \

@@ -23,7 +23,10 @@ setup(
             sources=["g2vec_amd/ops/csrc/bindings.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
-                "nvcc": ["-O3", "-std=c++17"],
+                # G2VEC_DEBUG=1: device bounds asserts + host debug info
+                "nvcc": ["-O3", "-std=c++17"] + (
+                    ["-DG2VEC_DEBUG=1", "-g"]
+                    if os.environ.get("G2VEC_DEBUG") == "1" else []),
             },
         )
     ],

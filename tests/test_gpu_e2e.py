@@ -74,3 +74,32 @@ def test_fast_equals_general_on_gpu(tmp_path):
         res = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None).train(ps)
         hists[path] = res.acc_val_history
     assert hists["fast"] == pytest.approx(hists["general"], abs=1e-5)
+
+
+@pytest.mark.timeout(600)
+def test_hipgraph_matches_eager(tmp_path):
+    """Graph-captured epochs must reproduce the eager trajectory exactly."""
+    import numpy as np
+    from g2vec_amd.models.cbow import CbowTrainer
+    from g2vec_amd.paths import PathSet
+    rng = np.random.default_rng(11)
+    G, P = 300, 600
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 20))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda", 0)
+    ps = PathSet(torch.tensor(genes, dtype=torch.int32, device=dev),
+                 torch.tensor(offs, dtype=torch.int32, device=dev),
+                 torch.tensor(labels, device=dev), G)
+    hists, weights = {}, {}
+    for graphed in (True, False):
+        cfg = G2VecConfig(hidden=128, epochs=10, early_stop=False, seed=3,
+                          device="cuda", use_hipgraph=graphed)
+        res = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None).train(ps)
+        hists[graphed] = res.acc_val_history
+        weights[graphed] = res.W_ih
+    assert hists[True] == hists[False]
+    assert torch.equal(weights[True], weights[False])

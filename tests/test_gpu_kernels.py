@@ -144,13 +144,17 @@ def test_adam_kernels_match_oracle():
 
 
 @pytest.mark.parametrize("hidden", [64, 128, 256])
-@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+@pytest.mark.parametrize("dtype", ["fp32", "bf16", "fp16"])
 def test_cbow_fwd_general_matches_oracle(hidden, dtype):
     genes, offs, labels = _pathset_tensors(seed=4)
     torch.manual_seed(1)
     W = torch.randn(300, hidden)
     who = torch.randn(hidden)
-    Wd = W.to(DEV).bfloat16().contiguous() if dtype == "bf16" else W.to(DEV)
+    Wd = W.to(DEV)
+    if dtype == "bf16":
+        Wd = Wd.bfloat16().contiguous()
+    elif dtype == "fp16":
+        Wd = Wd.half().contiguous()
     Wc = Wd.cpu().float()   # oracle sees the same (possibly rounded) weights
     lc, cc, dc, Hc = cpu_ref.cbow_fwd(Wc, who, genes, offs, labels, 1 / 500, True)
     lg, cg, dg, Hg = ops.cbow_fwd(Wd, who.to(DEV), genes.to(DEV), offs.to(DEV),
