@@ -158,9 +158,13 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
 
     for (int step = 0; step < len_path; ++step) {
       G2V_ASSERT(plen < len_path && cur >= 0);
-      if (lane == 0) vis[plen] = cur;
+      if (lane == 0) {
+        vis[plen] = cur;
+        // ONE lane inserts: a 64-lane same-address LDS store serializes its
+        // lane group and the LDS array is shared by all 32 waves of the CU
+        hset_insert(tab, tmask, (uint32_t)cur);
+      }
       ++plen;
-      hset_insert(tab, tmask, (uint32_t)cur);  // every lane: same probe, same write
       hash += gene_hash_dev((uint32_t)cur);
       const int s = row_ptr[cur], e = row_ptr[cur + 1];
       const int deg = e - s;
