@@ -133,9 +133,14 @@ __device__ __forceinline__ bool hset_contains(const uint32_t* tab,
   }
 }
 
+// Neighbors are packed {col, weight} pairs (one dwordx2 load per candidate
+// instead of two dword loads); rows with <= 4*64 neighbors sample entirely
+// from registers (weights cached across the total/selection passes).
+#define WCHUNKS 4
+
 extern "C" __global__ void __launch_bounds__(256)
-walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
-            const float* __restrict__ wgt, const int* __restrict__ sources,
+walk_kernel(const int* __restrict__ row_ptr, const int2* __restrict__ nbr,
+            const int* __restrict__ sources,
             int n_src, long long n_walks, int len_path, int tsize,
             uint64_t seed, int* __restrict__ out_nodes,
             int* __restrict__ out_len, long long* __restrict__ out_hash) {
@@ -170,38 +175,68 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
       const int deg = e - s;
       if (deg <= 0) break;
 
-      if (deg <= WAVE) {
-        // fast path: one candidate per lane, fully register-resident
-        int cand = -1;
-        float w = 0.f;
-        if (lane < deg) {
-          cand = col_idx[s + lane];
-          w = wgt[s + lane];
-          if (hset_contains(tab, tmask, (uint32_t)cand)) w = 0.f;
+      if (deg <= WCHUNKS * WAVE) {
+        // register path: the whole row (<= 4 chunks of 64) is loaded once,
+        // membership-masked once, and both the total and the selection use
+        // the cached registers — one dwordx2 load + one hash probe per
+        // candidate per STEP, nothing re-read.
+        const int nchunk = (deg + WAVE - 1) >> 6;
+        float wreg[WCHUNKS];
+        int creg[WCHUNKS];
+        float partial = 0.f;
+#pragma unroll
+        for (int k = 0; k < WCHUNKS; ++k) {
+          wreg[k] = 0.f;
+          creg[k] = -1;
+          const int j = (k << 6) + lane;
+          if (k < nchunk && j < deg) {
+            const int2 cw = nbr[s + j];
+            creg[k] = cw.x;
+            float w = __int_as_float(cw.y);
+            if (hset_contains(tab, tmask, (uint32_t)cw.x)) w = 0.f;
+            wreg[k] = w;
+            partial += w;
+          }
         }
-        const float tot = wave_sum(w);
+        const float tot = wave_sum(partial);
         const uint64_t r = sm64_next(state);   // drawn even on dead end
         if (!(tot > 0.f)) break;
         const float target = (float)(u01_from(r) * (double)tot);
-        const float scan = wave_incl_scan(w);
-        const bool hit = (w > 0.f) && (scan > target) && (scan - w <= target);
-        const unsigned long long mh = __ballot(hit);
-        int lane_sel;
-        if (mh != 0ULL) {
-          lane_sel = __ffsll((long long)mh) - 1;      // first crossing lane
-        } else {
-          // rounding tail: target >= total running sum -> last unvisited
+        int chosen_cand = -1;
+        float base = 0.f;
+        unsigned long long any_pos = 0ULL;
+        int last_pos_cand = -1;
+        for (int k = 0; k < nchunk; ++k) {
+          const float w = wreg[k];
+          const float scan = wave_incl_scan(w);
+          const float chunk_tot = __shfl(scan, WAVE - 1);
+          const bool hit = (w > 0.f) && (base + scan > target) &&
+                           (base + scan - w <= target);
+          const unsigned long long m = __ballot(hit);
+          if (m != 0ULL) {
+            chosen_cand = __shfl(creg[k], __ffsll((long long)m) - 1);
+            break;
+          }
           const unsigned long long mp = __ballot(w > 0.f);
-          lane_sel = 63 - __clzll((long long)mp);
+          if (mp != 0ULL) {
+            any_pos = 1;
+            last_pos_cand = __shfl(creg[k], 63 - __clzll((long long)mp));
+          }
+          base += chunk_tot;
         }
-        cur = __shfl(cand, lane_sel);
+        if (chosen_cand < 0) {
+          // rounding tail: target >= running total -> last unvisited
+          if (!any_pos) break;                 // cannot happen when tot > 0
+          chosen_cand = last_pos_cand;
+        }
+        cur = chosen_cand;
       } else {
-        // chunked path for high-degree nodes
+        // chunked fallback for very-high-degree nodes (> 256 neighbors)
         float partial = 0.f;
         for (int j = lane; j < deg; j += WAVE) {
-          const int c2 = col_idx[s + j];
-          float w = wgt[s + j];
-          if (hset_contains(tab, tmask, (uint32_t)c2)) w = 0.f;
+          const int2 cw = nbr[s + j];
+          float w = __int_as_float(cw.y);
+          if (hset_contains(tab, tmask, (uint32_t)cw.x)) w = 0.f;
           partial += w;
         }
         const float tot = wave_sum(partial);
@@ -214,9 +249,9 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           const int j = j0 + lane;
           float w = 0.f;
           if (j < deg) {
-            const int c2 = col_idx[s + j];
-            w = wgt[s + j];
-            if (hset_contains(tab, tmask, (uint32_t)c2)) w = 0.f;
+            const int2 cw = nbr[s + j];
+            w = __int_as_float(cw.y);
+            if (hset_contains(tab, tmask, (uint32_t)cw.x)) w = 0.f;
           }
           const float scan = wave_incl_scan(w);
           const float chunk_tot = __shfl(scan, WAVE - 1);
@@ -232,16 +267,16 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             const int j = j0 + lane;
             float w = 0.f;
             if (j < deg) {
-              const int c2 = col_idx[s + j];
-              w = wgt[s + j];
-              if (hset_contains(tab, tmask, (uint32_t)c2)) w = 0.f;
+              const int2 cw = nbr[s + j];
+              w = __int_as_float(cw.y);
+              if (hset_contains(tab, tmask, (uint32_t)cw.x)) w = 0.f;
             }
             const unsigned long long m = __ballot(w > 0.f);
             if (m != 0ULL) chosen = j0 + (63 - __clzll((long long)m));
           }
           if (chosen < 0) break;               // cannot happen when tot > 0
         }
-        cur = col_idx[s + chosen];
+        cur = nbr[s + chosen].x;
       }
     }
 
