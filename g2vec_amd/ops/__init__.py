@@ -36,18 +36,10 @@ def native():
 
 
 # ------------------------------------------------------------------ walks
-def pack_neighbors(col_idx: torch.Tensor, weights: torch.Tensor) -> torch.Tensor:
-    """Interleaved i32 [nnz, 2] {col, weight-bits}: the walk kernel reads
-    one dwordx2 per neighbor instead of two dword loads."""
-    return torch.stack([col_idx, weights.view(torch.int32)], dim=1).contiguous()
-
-
 def random_walks(row_ptr, col_idx, weights, sources, num_repetition: int,
-                 len_path: int, seed: int, packed: Optional[torch.Tensor] = None):
+                 len_path: int, seed: int):
     if row_ptr.is_cuda:
-        if packed is None:
-            packed = pack_neighbors(col_idx, weights)
-        return native().random_walks(row_ptr, packed, sources,
+        return native().random_walks(row_ptr, col_idx, weights, sources,
                                      num_repetition, len_path, seed)
     return cpu_ref.random_walks(row_ptr, col_idx, weights, sources,
                                 num_repetition, len_path, seed)

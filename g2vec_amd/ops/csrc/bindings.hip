@@ -39,15 +39,13 @@ inline int grid_for(long long work_items, int per_block) {
   } while (0)
 
 // ------------------------------------------------------------------ walks
-std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor packed,
-                                        torch::Tensor sources,
+std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col_idx,
+                                        torch::Tensor weights, torch::Tensor sources,
                                         int64_t num_repetition, int64_t len_path,
                                         int64_t seed) {
-  // packed: i32 [nnz, 2] interleaved {col, weight-bits} (one dwordx2 per
-  // neighbor in the kernel)
   CHECK_DEV(row_ptr); CHECK_CONT(row_ptr); CHECK_I32(row_ptr);
-  CHECK_DEV(packed); CHECK_CONT(packed); CHECK_I32(packed);
-  TORCH_CHECK(packed.dim() == 2 && packed.size(1) == 2, "packed must be [nnz,2]");
+  CHECK_DEV(col_idx); CHECK_CONT(col_idx); CHECK_I32(col_idx);
+  CHECK_DEV(weights); CHECK_CONT(weights); CHECK_F32(weights);
   CHECK_DEV(sources); CHECK_CONT(sources); CHECK_I32(sources);
   TORCH_CHECK(len_path >= 1 && len_path <= 512, "len_path out of range");
   const long long n_src = sources.numel();
@@ -64,7 +62,7 @@ std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor pac
   const size_t lds = (size_t)wpb * (len_path + tsize) * sizeof(int);
   hipLaunchKernelGGL(walk_kernel, dim3(grid_for(n_walks, wpb)), dim3(256), lds,
                      cur_stream(), row_ptr.data_ptr<int>(),
-                     (const int2*)packed.data_ptr<int>(),
+                     col_idx.data_ptr<int>(), weights.data_ptr<float>(),
                      sources.data_ptr<int>(),
                      (int)n_src, n_walks, (int)len_path, tsize, (uint64_t)seed,
                      nodes.data_ptr<int>(), lengths.data_ptr<int>(),
