@@ -76,11 +76,16 @@ def random_walks(row_ptr: torch.Tensor, col_idx: torch.Tensor, weights: torch.Te
     hashes = np.zeros(n_walks, dtype=np.int64)
 
     for wid in range(n_walks):
+        rep = wid // n_src
+        source = int(src[wid % n_src])
+        # RNG keyed on the GLOBAL (source, repetition) so DP-sharded
+        # generation is bitwise-identical to single-process
+        gid = np.uint64(source) * np.uint64(num_repetition) + np.uint64(rep)
         with np.errstate(over="ignore"):
-            state = np.uint64(np.uint64(seed) ^ np.uint64(np.uint64(wid) * _SM64_M2 + np.uint64(1)))
+            state = np.uint64(np.uint64(seed) ^ np.uint64(gid * _SM64_M2 + np.uint64(1)))
         # warm the stream like the device kernel does
         state, _ = splitmix64(state)
-        cur = int(src[wid % n_src])
+        cur = source
         visited = []
         vset = set()
         for _step in range(len_path):

@@ -64,7 +64,8 @@ std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col
                      cur_stream(), row_ptr.data_ptr<int>(),
                      col_idx.data_ptr<int>(), weights.data_ptr<float>(),
                      sources.data_ptr<int>(),
-                     (int)n_src, n_walks, (int)len_path, tsize, (uint64_t)seed,
+                     (int)n_src, n_walks, (int)num_repetition, (int)len_path,
+                     tsize, (uint64_t)seed,
                      nodes.data_ptr<int>(), lengths.data_ptr<int>(),
                      (long long*)hashes.data_ptr<int64_t>());
   LAUNCH_CHECK();

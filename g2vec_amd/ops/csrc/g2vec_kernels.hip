@@ -143,7 +143,7 @@ __device__ __forceinline__ bool hset_contains(const uint32_t* tab,
 extern "C" __global__ void __launch_bounds__(256)
 walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             const float* __restrict__ wgt, const int* __restrict__ sources,
-            int n_src, long long n_walks, int len_path, int tsize,
+            int n_src, long long n_walks, int num_rep, int len_path, int tsize,
             uint64_t seed, int* __restrict__ out_nodes,
             int* __restrict__ out_len, long long* __restrict__ out_hash) {
   extern __shared__ int smem[];
@@ -156,9 +156,14 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
 
   for (long long walk = (long long)blockIdx.x * wpb + wib; walk < n_walks;
        walk += (long long)gridDim.x * wpb) {
-    uint64_t state = seed ^ (uint64_t)((uint64_t)walk * 0x94D049BB133111EBULL + 1ULL);
+    const int rep = (int)(walk / n_src);
+    const int src = sources[walk % n_src];
+    // RNG keyed on the GLOBAL (source, repetition): DP-sharded generation
+    // is bitwise-identical to single-process (C5 rank invariance)
+    const uint64_t gid = (uint64_t)src * (uint64_t)num_rep + (uint64_t)rep;
+    uint64_t state = seed ^ (uint64_t)(gid * 0x94D049BB133111EBULL + 1ULL);
     (void)sm64_next(state);                    // warm draw (CPU oracle parity)
-    int cur = sources[walk % n_src];
+    int cur = src;
     int plen = 0;
     uint64_t hash = 0;
     for (int i = lane; i < tsize; i += WAVE) tab[i] = HSET_EMPTY;

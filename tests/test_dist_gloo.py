@@ -83,3 +83,53 @@ def test_dp2_matches_single_process():
     res = tr.train(ps)
     assert hist_dp == pytest.approx(res.acc_val_history, abs=1e-6)
     assert np.allclose(W_dp, res.W_ih.numpy(), atol=1e-5)
+
+
+def _pipeline_worker(rank, world, port, files, outdir, q):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        from g2vec_amd.pipeline import run
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        cfg = G2VecConfig(expression_file=files["expression"],
+                          clinical_file=files["clinical"],
+                          network_file=files["network"],
+                          result_name=f"{outdir}/dp{rank}",
+                          len_path=12, num_repetition=2, epochs=8,
+                          device="cpu", seed=0, early_stop=False)
+        res = run(cfg, ctx)
+        if rank == 0:
+            q.put((res["n_paths"], res["n_genes_in_paths"], res["acc_val"]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_pipeline_end_to_end(tiny_files, tmp_path):
+    """Full pipeline under DP=2 (gloo): the sharded walk generation +
+    all-gather must reproduce the single-process path set EXACTLY (the walk
+    RNG is keyed on the global (source, repetition) pair)."""
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_pipeline_worker,
+                          args=(r, 2, port, tiny_files, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    n_paths_dp, n_gip_dp, acc_dp = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    from g2vec_amd.pipeline import run
+    cfg = G2VecConfig(expression_file=tiny_files["expression"],
+                      clinical_file=tiny_files["clinical"],
+                      network_file=tiny_files["network"],
+                      result_name=str(tmp_path / "sp"),
+                      len_path=12, num_repetition=2, epochs=8,
+                      device="cpu", seed=0, early_stop=False)
+    res = run(cfg)
+    assert n_paths_dp == res["n_paths"]
+    assert n_gip_dp == res["n_genes_in_paths"]
+    assert abs(acc_dp - res["acc_val"]) < 0.05
