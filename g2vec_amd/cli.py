@@ -47,6 +47,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph capture of the training epoch")
     p.add_argument("--save-paths", type=str, default="")
+    p.add_argument("--save-model", type=str, default="",
+                   help="save trained weights + metadata as a .pt checkpoint")
     p.add_argument("--load-paths", type=str, default="")
     p.add_argument("--log-jsonl", type=str, default="")
     return p
@@ -63,7 +65,8 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         pcc_mode=a.pcc_mode, trainer_path=a.trainer_path,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, save_paths=a.save_paths,
-        load_paths=a.load_paths, log_jsonl=a.log_jsonl,
+        load_paths=a.load_paths, save_model=a.save_model,
+        log_jsonl=a.log_jsonl,
         use_hipgraph=not a.no_hipgraph)
 
 
@@ -71,7 +74,12 @@ def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
     cfg = args_to_config(args)
     ctx = init_dist(cfg.device)
-    run(cfg, ctx)
+    try:
+        run(cfg, ctx)
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
     return 0
 
 

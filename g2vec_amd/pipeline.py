@@ -122,6 +122,13 @@ def run(cfg: G2VecConfig, ctx: Optional[DistContext] = None) -> Dict:
     jsonl.emit("train", acc_val=res.acc_val, acc_tr=res.acc_tr,
                stop_epoch=res.stop_epoch, epochs_run=res.epochs_run,
                wall_to_acc088_s=res.wall_to_acc_s)
+    if cfg.save_model and ctx.is_primary:
+        torch.save({"W_ih": res.W_ih.float().cpu(),
+                    "acc_val": res.acc_val, "acc_tr": res.acc_tr,
+                    "stop_epoch": res.stop_epoch,
+                    "hidden": cfg.hidden, "n_genes": n_genes,
+                    "gene_index": list(map(str, data["gene"]))},
+                   cfg.save_model)
 
     W = res.W_ih.float().cpu().numpy()
     result: Dict = {
