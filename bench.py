@@ -84,6 +84,7 @@ def main() -> int:
     ap.add_argument("--n-extra", type=int, default=2381)
     ap.add_argument("--n-modules", type=int, default=16)
     ap.add_argument("--acc-target-epochs", type=int, default=60)
+    ap.add_argument("--no-hipgraph", action="store_true")
     args = ap.parse_args()
 
     ctx = init_dist("auto")
@@ -99,7 +100,8 @@ def main() -> int:
 
     cfg = G2VecConfig(hidden=args.hidden, len_path=args.len_path,
                       num_repetition=args.reps, epochs=500, seed=args.seed,
-                      device=str(device.type), trainer_path=args.trainer_path)
+                      device=str(device.type), trainer_path=args.trainer_path,
+                      use_hipgraph=not args.no_hipgraph)
 
     # ---- dataset + graphs + walks (input pipeline; measured, not the metric)
     expr, labels, edge_idx, n_genes = build_dataset(
@@ -186,6 +188,7 @@ def main() -> int:
                 "hidden": args.hidden,
                 "num_repetition": args.reps,
                 "trainer_path": args.trainer_path,
+                "hipgraph_active": bool(getattr(st, "graph", None) is not None),
                 "val_acc": round(acc_val, 4),
                 "wall_to_val_acc_0.88_s": (round(wall_to_acc, 4)
                                            if wall_to_acc else None),
