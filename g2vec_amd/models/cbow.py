@@ -145,6 +145,15 @@ class CbowTrainer:
         st.counts_buf = torch.zeros(2, dtype=torch.float32, device=self.device)
         st.graph = None
         st.graph_failed = False
+        # concatenated train+val evaluation set: both accuracy splits ride
+        # ONE forward kernel per epoch
+        if not use_general and tr.n_paths + vl.n_paths > 0:
+            off_vl = vl.offsets.int() + tr.offsets[-1]
+            st.ev_genes = torch.cat([tr.genes, vl.genes]).contiguous()
+            st.ev_offsets = torch.cat([tr.offsets, off_vl[1:]]).contiguous()
+            st.ev_labels = torch.cat([tr.labels, vl.labels]).contiguous()
+        else:
+            st.ev_genes = None
         return st
 
     def _epoch_body_fast(self, st) -> None:
@@ -165,14 +174,14 @@ class CbowTrainer:
                        self.B1, self.B2, self.EPS, lrt_buf=lrt)
         ops.adam_dense(st.who, st.mO, st.vO, grad_who, st.t_adam, cfg.lr,
                        self.B1, self.B2, self.EPS, lrt_buf=lrt)
-        # post-update accuracy (reference order, G2Vec.py:264-267)
+        # post-update accuracy (reference order, G2Vec.py:264-267):
+        # one forward over the concatenated train+val paths
         torch.mv(st.W, st.who, out=st.s_buf)
-        for k, split in enumerate((tr, vl)):
-            if split.n_paths == 0:
-                continue
+        if st.ev_genes is not None:
             _l, corr, _d = ops.cbow_fwd_scalar(
-                st.s_buf, split.genes, split.offsets, split.labels, 1.0, False)
-            st.counts_buf[k].copy_(corr.sum())
+                st.s_buf, st.ev_genes, st.ev_offsets, st.ev_labels, 1.0, False)
+            st.counts_buf[0].copy_(corr[:tr.n_paths].sum())
+            st.counts_buf[1].copy_(corr[tr.n_paths:].sum())
         self.ctx.allreduce_(st.counts_buf)      # C3: one fused metric reduce
 
     def run_epoch(self, st) -> tuple:
