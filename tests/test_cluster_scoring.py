@@ -81,3 +81,16 @@ def test_tscores_shape():
     labels = np.array([0] * 12 + [1] * 13)
     ts = t_scores(expr, labels)
     assert ts.shape == (6,) and (ts >= 0).all()
+
+
+def test_t_scores_vectorized_matches_scalar_oracle():
+    rng = np.random.default_rng(9)
+    expr = rng.normal(size=(35, 40)).astype(np.float32)
+    expr[:, 7] = 1.0                      # zero-variance column -> t == 0
+    labels = np.array([0] * 18 + [1] * 17)
+    ts = t_scores(expr, labels)
+    good, poor = expr[labels == 0], expr[labels == 1]
+    for i in range(expr.shape[1]):
+        ref = abs(t_statistic(good[:, i].astype(np.float64),
+                              poor[:, i].astype(np.float64)))
+        assert abs(ts[i] - ref) < 1e-5, i
