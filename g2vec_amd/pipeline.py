@@ -143,7 +143,8 @@ def run(cfg: G2VecConfig, ctx: Optional[DistContext] = None) -> Dict:
 
     log(">>> 5. Find L-groups")
     with timers.phase("lgroups"):
-        lg = find_lgroups(W, freq.cpu().numpy(), cfg.compat_lgroup_bug)
+        lg = find_lgroups(W, freq.cpu().numpy(), cfg.compat_lgroup_bug,
+                          device=device)
 
     log(">>> 6. Select biomarkers with gene scores")
     with timers.phase("scoring"):

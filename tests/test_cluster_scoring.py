@@ -94,3 +94,15 @@ def test_t_scores_vectorized_matches_scalar_oracle():
         ref = abs(t_statistic(good[:, i].astype(np.float64),
                               poor[:, i].astype(np.float64)))
         assert abs(ts[i] - ref) < 1e-5, i
+
+
+def test_kmeans_torch_backend_matches_sklearn_grouping():
+    emb = _embeddings()
+    freq = np.full(150, 2, dtype=np.int64)
+    freq[:20] = 0
+    freq[20:50] = 1
+    lg_sk = find_lgroups(emb, freq, backend="sklearn")
+    lg_t = find_lgroups(emb, freq, backend="torch")
+    # well-separated clusters: identical L-group assignment after the
+    # size/frequency-based relabelling
+    assert (lg_sk == lg_t).all()
