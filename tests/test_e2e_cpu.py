@@ -81,3 +81,20 @@ def test_jsonl_metrics(tiny_files, tmp_path):
     events = [json.loads(l) for l in open(log)]
     names = {e["event"] for e in events}
     assert {"counts", "paths", "train", "phase"} <= names
+
+
+def test_unseeded_mode_runs(tiny_files, tmp_path):
+    """seed=None reproduces the reference's nondeterministic behaviour
+    without crashing (manual.pdf p.4 documents run-to-run variation)."""
+    res = run(_cfg(tiny_files, tmp_path, seed=None, epochs=5,
+                   num_repetition=1))
+    assert res["n_paths"] > 0
+
+
+def test_missing_file_raises(tmp_path):
+    cfg = G2VecConfig(expression_file=str(tmp_path / "nope.tsv"),
+                      clinical_file=str(tmp_path / "nope2.tsv"),
+                      network_file=str(tmp_path / "nope3.tsv"),
+                      result_name=str(tmp_path / "out"), device="cpu")
+    with pytest.raises(Exception):
+        run(cfg)
