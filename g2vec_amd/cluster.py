@@ -55,9 +55,9 @@ def find_lgroups(embeddings: np.ndarray, gene_freq: np.ndarray,
     """embeddings: f32 [G, h]; gene_freq: int [G] in {0 good,1 poor,2 other}.
     Returns int32 [G] with 0 good / 1 poor / 2 other.
     backend: 'sklearn' (reference parity: KMeans random_state=0) |
-    'torch' (scalable, GPU-capable) | 'auto' (sklearn below 200k genes)."""
+    'torch' (scalable, GPU-capable) | 'auto' (sklearn up to 50k genes)."""
     if backend == "auto":
-        backend = "sklearn" if embeddings.shape[0] <= 200_000 else "torch"
+        backend = "sklearn" if embeddings.shape[0] <= 50_000 else "torch"
     if backend == "torch":
         import torch
         X = torch.from_numpy(np.ascontiguousarray(embeddings))

@@ -61,3 +61,13 @@ def test_writers_golden(tmp_path):
     write_vectors(base, mat, ["A1CF"])
     assert (tmp_path / "res_vectors.txt").read_text() == (
         "GeneSymbol\tV0\tV1\nA1CF\t0.123457\t-1.000000\n")
+
+
+def test_write_vectors_engines_identical(tmp_path):
+    rng = np.random.default_rng(4)
+    mat = rng.normal(size=(7, 5)).astype(np.float32)
+    genes = [f"G{i}" for i in range(7)]
+    from g2vec_amd.io.writers import write_vectors
+    a = write_vectors(str(tmp_path / "a"), mat, genes, engine="numpy")
+    b = write_vectors(str(tmp_path / "b"), mat, genes, engine="pandas")
+    assert open(a).read() == open(b).read()
