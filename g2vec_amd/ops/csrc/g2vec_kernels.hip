@@ -59,9 +59,25 @@ __device__ __forceinline__ uint64_t gene_hash_dev(uint32_t g) {
 }
 
 // ---------------------------------------------------------------- wave utils
+__device__ __forceinline__ int uni(int v) {
+  return __builtin_amdgcn_readfirstlane(v);
+}
+
+__device__ __forceinline__ float unif(float v) {
+  return __int_as_float(__builtin_amdgcn_readfirstlane(__float_as_int(v)));
+}
+
 __device__ __forceinline__ float wave_sum(float v) {
   for (int o = 32; o; o >>= 1) v += __shfl_down(v, o);
   return __shfl(v, 0);
+}
+
+// wave reduction with a PROVABLY-uniform (SGPR) result: downstream
+// branches compile to scalar branches instead of exec-mask dances, and
+// dependent integer math lands on the free scalar pipe
+__device__ __forceinline__ float wave_sum_uni(float v) {
+  for (int o = 32; o; o >>= 1) v += __shfl_down(v, o);
+  return unif(v);   // lane 0 holds the total; full EXEC here
 }
 
 __device__ __forceinline__ float wave_incl_scan(float v) {
@@ -148,7 +164,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             int* __restrict__ out_len, long long* __restrict__ out_hash) {
   extern __shared__ int smem[];
   const int lane = threadIdx.x & (WAVE - 1);
-  const int wib = threadIdx.x >> 6;            // wave-in-block
+  const int wib = uni(threadIdx.x >> 6);       // wave-in-block (SGPR)
   const int wpb = blockDim.x >> 6;             // waves per block
   int* vis = smem + wib * len_path;            // ordered path (output)
   uint32_t* tab = (uint32_t*)(smem + wpb * len_path) + wib * tsize;
@@ -157,7 +173,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
   for (long long walk = (long long)blockIdx.x * wpb + wib; walk < n_walks;
        walk += (long long)gridDim.x * wpb) {
     const int rep = (int)(walk / n_src);
-    const int src = sources[walk % n_src];
+    const int src = uni(sources[walk % n_src]);
     // RNG keyed on the GLOBAL (source, repetition): DP-sharded generation
     // is bitwise-identical to single-process (C5 rank invariance)
     const uint64_t gid = (uint64_t)src * (uint64_t)num_rep + (uint64_t)rep;
@@ -192,7 +208,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           w = wgt[s + j];
           if (hset_contains(tab, tmask, (uint32_t)cand)) w = 0.f;
         }
-        const float tot = wave_sum(w);
+        const float tot = wave_sum_uni(w);
         const uint64_t r = sm64_next(state);   // drawn even on dead end
         if (!(tot > 0.f)) break;
         const float target = (float)(u01_from(r) * (double)tot);
@@ -206,7 +222,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           const unsigned long long mp = __ballot(w > 0.f);
           lane_sel = 63 - __clzll((long long)mp);
         }
-        cur = __shfl(cand, lane_sel);
+        cur = uni(__shfl(cand, lane_sel));
       } else if (deg <= WCHUNKS * WAVE) {
         // register path: the whole row (<= 4 chunks of 64) is loaded once,
         // membership-masked once, and both the total and the selection use
@@ -230,7 +246,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             partial += w;
           }
         }
-        const float tot = wave_sum(partial);
+        const float tot = wave_sum_uni(partial);
         const uint64_t r = sm64_next(state);   // drawn even on dead end
         if (!(tot > 0.f)) break;
         const float target = (float)(u01_from(r) * (double)tot);
@@ -241,18 +257,18 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
         for (int k = 0; k < nchunk; ++k) {
           const float w = wreg[k];
           const float scan = wave_incl_scan(w);
-          const float chunk_tot = __shfl(scan, WAVE - 1);
+          const float chunk_tot = unif(__shfl(scan, WAVE - 1));
           const bool hit = (w > 0.f) && (base + scan > target) &&
                            (base + scan - w <= target);
           const unsigned long long m = __ballot(hit);
           if (m != 0ULL) {
-            chosen_cand = __shfl(creg[k], __ffsll((long long)m) - 1);
+            chosen_cand = uni(__shfl(creg[k], __ffsll((long long)m) - 1));
             break;
           }
           const unsigned long long mp = __ballot(w > 0.f);
           if (mp != 0ULL) {
             any_pos = 1;
-            last_pos_cand = __shfl(creg[k], 63 - __clzll((long long)mp));
+            last_pos_cand = uni(__shfl(creg[k], 63 - __clzll((long long)mp)));
           }
           base += chunk_tot;
         }
@@ -271,7 +287,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
           partial += w;
         }
-        const float tot = wave_sum(partial);
+        const float tot = wave_sum_uni(partial);
         const uint64_t r = sm64_next(state);
         if (!(tot > 0.f)) break;
         const float target = (float)(u01_from(r) * (double)tot);
@@ -286,7 +302,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
           }
           const float scan = wave_incl_scan(w);
-          const float chunk_tot = __shfl(scan, WAVE - 1);
+          const float chunk_tot = unif(__shfl(scan, WAVE - 1));
           const bool hit = (j < deg) && (w > 0.f) &&
                            (base + scan > target) && (base + scan - w <= target);
           const unsigned long long m = __ballot(hit);
@@ -308,7 +324,7 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           }
           if (chosen < 0) break;               // cannot happen when tot > 0
         }
-        cur = col_idx[s + chosen];
+        cur = uni(col_idx[s + chosen]);
       }
     }
 
