@@ -208,11 +208,13 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           w = wgt[s + j];
           if (hset_contains(tab, tmask, (uint32_t)cand)) w = 0.f;
         }
-        const float tot = wave_sum_uni(w);
+        // one scan serves both the total (its lane-63 element) and the
+        // selection — 6 fewer dependent ds_bpermute per step
+        const float scan = wave_incl_scan(w);
+        const float tot = unif(__shfl(scan, WAVE - 1));
         const uint64_t r = sm64_next(state);   // drawn even on dead end
         if (!(tot > 0.f)) break;
         const float target = (float)(u01_from(r) * (double)tot);
-        const float scan = wave_incl_scan(w);
         const bool hit = (w > 0.f) && (scan > target) && (scan - w <= target);
         const unsigned long long mh = __ballot(hit);
         int lane_sel;
