@@ -130,9 +130,9 @@ class CbowTrainer:
         st.W16 = None
         if use_general and cfg.dtype != "fp32":
             st.W16 = (W.bfloat16() if cfg.dtype == "bf16" else W.half())
-        st.plan = None
-        if not use_general:
-            st.plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)
+        # gene-sorted instance plan: drives the deterministic backward on
+        # both trainer paths (genes never change across epochs)
+        st.plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)
         st.inv_b = 1.0 / max(self.n_tr_global, 1)
         P_loc = tr.n_paths
         bs = cfg.batch_size if cfg.batch_size > 0 else P_loc
@@ -226,8 +226,7 @@ class CbowTrainer:
                 (hi - lo) * self.ctx.world)
             st.t_adam += 1
             if cfg.trainer_path == "general":
-                self._step_general(st.W, st.W16, st.who, st.mW, st.vW, st.mO,
-                                   st.vO, st.tr, lo, hi, b_inv, st.t_adam)
+                self._step_general(st, lo, hi, b_inv, st.t_adam)
             else:
                 self._step_fast(st, lo, hi, b_inv, st.t_adam)
         if cfg.trainer_path == "general":
@@ -323,12 +322,16 @@ class CbowTrainer:
         ops.adam_dense(who, st.mO, st.vO, grad_who, t, self.cfg.lr, self.B1,
                        self.B2, self.EPS)
 
-    def _step_general(self, W, W16, who, mW, vW, mO, vO, tr, lo, hi, inv_b, t):
+    def _step_general(self, st, lo, hi, inv_b, t):
+        W, W16, who, tr = st.W, st.W16, st.who, st.tr
+        mW, vW, mO, vO = st.mW, st.vW, st.mO, st.vO
         genes, offsets, labels = self._slice(tr, lo, hi)
         Wg = W16 if W16 is not None else W
         _loss, _corr, dO, H = ops.cbow_fwd(Wg, who, genes, offsets, labels,
                                            inv_b, True)
-        dW = ops.cbow_bwd_rows(who, genes, offsets, dO, self.G)
+        full = lo == 0 and hi == tr.n_paths
+        dW = ops.cbow_bwd_rows(who, genes, offsets, dO, self.G,
+                               plan=(st.plan if full else None))
         grad_who = torch.mv(H.t(), dO)
         self.ctx.allreduce_(dW)
         self.ctx.allreduce_(grad_who)

@@ -130,9 +130,16 @@ def cbow_fwd(W, who, genes, offsets, labels, inv_b: float, want_grad: bool):
     return cpu_ref.cbow_fwd(W, who, genes, offsets, labels, inv_b, want_grad)
 
 
-def cbow_bwd_rows(who, genes, offsets, dO, n_genes: int):
+def cbow_bwd_rows(who, genes, offsets, dO, n_genes: int,
+                  plan: Optional[ScatterPlan] = None):
+    """General-path dW_ih backward. On GPU this is the deterministic
+    per-gene-segment kernel (no atomics) driven by the same ScatterPlan as
+    the fast path's c-reduction."""
     if dO.is_cuda:
-        return native().cbow_bwd_rows(who, genes, offsets, dO, int(n_genes))
+        if plan is None:
+            plan = build_scatter_plan(genes, offsets, n_genes)
+        return native().cbow_bwd_rows(who, plan.inst_path, plan.seg_start,
+                                      plan.seg_gene, dO, int(n_genes))
     return cpu_ref.cbow_bwd_rows(who, genes, offsets, dO, n_genes)
 
 

@@ -216,27 +216,29 @@ std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
   return {loss, correct, dO, H};
 }
 
-torch::Tensor cbow_bwd_rows(torch::Tensor who, torch::Tensor genes,
-                            torch::Tensor offs, torch::Tensor dO,
-                            int64_t n_genes) {
+torch::Tensor cbow_bwd_rows(torch::Tensor who, torch::Tensor inst_path,
+                            torch::Tensor seg_start, torch::Tensor seg_gene,
+                            torch::Tensor dO, int64_t n_genes) {
   CHECK_DEV(who); CHECK_CONT(who); CHECK_F32(who);
-  CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
-  CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
+  CHECK_DEV(inst_path); CHECK_CONT(inst_path); CHECK_I32(inst_path);
+  CHECK_DEV(seg_start); CHECK_CONT(seg_start); CHECK_I32(seg_start);
+  CHECK_DEV(seg_gene); CHECK_CONT(seg_gene); CHECK_I32(seg_gene);
   CHECK_DEV(dO); CHECK_CONT(dO); CHECK_F32(dO);
   const int h = (int)who.numel();
   const int hpl = h / 64;
   TORCH_CHECK(h % 64 == 0 && hpl >= 1 && hpl <= 16 &&
               (hpl & (hpl - 1)) == 0, "hidden must be 64*{1,2,4,8,16}");
-  const long long P = dO.numel();
   auto dW = torch::zeros({n_genes, h},
                          torch::TensorOptions().dtype(at::kFloat).device(dO.device()));
-  if (P == 0) return dW;
-  const int grid = grid_for(P, 4);
+  const int n_seg = (int)seg_gene.numel();
+  if (n_seg == 0) return dW;
+  const int grid = grid_for(n_seg, 4);
 #define BWD_CASE(HPL)                                                         \
-  hipLaunchKernelGGL((cbow_bwd_rows_kernel<HPL>), dim3(grid), dim3(256), 0,   \
-                     cur_stream(), who.data_ptr<float>(),                     \
-                     genes.data_ptr<int>(), offs.data_ptr<int>(),             \
-                     dO.data_ptr<float>(), P, h, dW.data_ptr<float>())
+  hipLaunchKernelGGL((cbow_bwd_rows_det_kernel<HPL>), dim3(grid), dim3(256),  \
+                     0, cur_stream(), who.data_ptr<float>(),                  \
+                     inst_path.data_ptr<int>(), seg_start.data_ptr<int>(),    \
+                     seg_gene.data_ptr<int>(), n_seg,                         \
+                     dO.data_ptr<float>(), h, dW.data_ptr<float>())
   switch (hpl) {
     case 1: BWD_CASE(1); break;
     case 2: BWD_CASE(2); break;
