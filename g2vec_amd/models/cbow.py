@@ -175,13 +175,11 @@ class CbowTrainer:
         ops.adam_dense(st.who, st.mO, st.vO, grad_who, st.t_adam, cfg.lr,
                        self.B1, self.B2, self.EPS, lrt_buf=lrt)
         # post-update accuracy (reference order, G2Vec.py:264-267):
-        # one forward over the concatenated train+val paths
+        # one fused eval kernel over the concatenated train+val paths
         torch.mv(st.W, st.who, out=st.s_buf)
         if st.ev_genes is not None:
-            _l, corr, _d = ops.cbow_fwd_scalar(
-                st.s_buf, st.ev_genes, st.ev_offsets, st.ev_labels, 1.0, False)
-            st.counts_buf[0].copy_(corr[:tr.n_paths].sum())
-            st.counts_buf[1].copy_(corr[tr.n_paths:].sum())
+            ops.cbow_eval_counts_(st.s_buf, st.ev_genes, st.ev_offsets,
+                                  st.ev_labels, tr.n_paths, st.counts_buf)
         self.ctx.allreduce_(st.counts_buf)      # C3: one fused metric reduce
 
     def run_epoch(self, st) -> tuple:

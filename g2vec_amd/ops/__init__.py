@@ -54,6 +54,20 @@ def cbow_fwd_scalar(s, genes, offsets, labels, inv_b: float, want_grad: bool):
     return cpu_ref.cbow_fwd_scalar(s, genes, offsets, labels, inv_b, want_grad)
 
 
+def cbow_eval_counts_(s, genes, offsets, labels, p_split: int,
+                      counts: torch.Tensor) -> None:
+    """Accumulate the concatenated train+val correct counts into counts[2]
+    (caller zeroes it). One fused kernel on GPU; oracle math on CPU."""
+    if s.is_cuda:
+        native().cbow_eval_counts_(s, genes, offsets, labels, int(p_split),
+                                   counts)
+        return
+    _l, corr, _d = cpu_ref.cbow_fwd_scalar(s, genes, offsets, labels, 1.0,
+                                           False)
+    counts[0] += corr[:p_split].sum()
+    counts[1] += corr[p_split:].sum()
+
+
 class ScatterPlan(NamedTuple):
     """Precomputed gene-sorted instance layout for the deterministic
     c = X^T dO reduction (built once per path set; genes never change

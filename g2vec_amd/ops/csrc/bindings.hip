@@ -86,14 +86,33 @@ std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
   auto correct = torch::empty({P}, opts);
   auto dO = want_grad ? torch::empty({P}, opts) : torch::empty({0}, opts);
   if (P == 0) return {loss, correct, dO};
-  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid_for(P, 4)), dim3(256), 0,
-                     cur_stream(), s.data_ptr<float>(), genes.data_ptr<int>(),
+  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid_for(P, 256)), dim3(256),
+                     0, cur_stream(), s.data_ptr<float>(), genes.data_ptr<int>(),
                      offs.data_ptr<int>(), labels.data_ptr<float>(), P,
                      (float)inv_b, loss.data_ptr<float>(),
                      correct.data_ptr<float>(),
                      want_grad ? dO.data_ptr<float>() : nullptr);
   LAUNCH_CHECK();
   return {loss, correct, dO};
+}
+
+void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
+                       torch::Tensor labels, int64_t p_split,
+                       torch::Tensor counts) {
+  CHECK_DEV(s); CHECK_CONT(s); CHECK_F32(s);
+  CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
+  CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
+  CHECK_DEV(labels); CHECK_CONT(labels); CHECK_F32(labels);
+  CHECK_DEV(counts); CHECK_CONT(counts); CHECK_F32(counts);
+  TORCH_CHECK(counts.numel() == 2, "counts must have 2 elements");
+  const long long P = labels.numel();
+  if (P == 0) return;
+  hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid_for(P, 256)),
+                     dim3(256), 0, cur_stream(), s.data_ptr<float>(),
+                     genes.data_ptr<int>(), offs.data_ptr<int>(),
+                     labels.data_ptr<float>(), P, (long long)p_split,
+                     counts.data_ptr<float>());
+  LAUNCH_CHECK();
 }
 
 torch::Tensor scatter_dO_det(torch::Tensor inst_path, torch::Tensor seg_start,
@@ -354,6 +373,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("random_walks", &random_walks, "CSR biased random walks (gfx950)");
   m.def("cbow_fwd_scalar", &cbow_fwd_scalar, "scalar CBOW forward + loss");
   m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO");
+  m.def("cbow_eval_counts_", &cbow_eval_counts_,
+        "fused train/val correct-count eval (in-place counts[2])");
   m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");
   m.def("adam_dense", &adam_dense, "TF1 Adam, dense grad");
   m.def("cbow_fwd", &cbow_fwd, "row-gather CBOW forward");
