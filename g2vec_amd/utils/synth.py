@@ -20,59 +20,12 @@ within-module edges; inactive/cross-module PCC < 0.1 drops out.
 """
 from __future__ import annotations
 
-from collections import deque
 from typing import Dict, List, Sequence, Tuple
 
 import numpy as np
 
 ACTIVE_LOADING = 1.8
 INACTIVE_LOADING = 0.25
-
-
-def assign_modules_graph(genes: Sequence[str], edges, n_modules: int,
-                         seed: int, dead_frac: float = 0.0) -> np.ndarray:
-    """Graph-contiguous module ids via multi-seed BFS flood fill over the
-    undirected network, so within-module network edges exist in bulk.
-
-    A `dead_frac` fraction of genes is marked module -1 ("dead": no factor
-    loading, pure noise expression -> isolated in BOTH group graphs). Their
-    single-node walks appear in both groups and are removed as common paths,
-    reproducing the reference dataset's structure where only ~half the
-    common genes appear in any path (README.md:32: 3,773 of 7,523)."""
-    g2i = {g: i for i, g in enumerate(genes)}
-    n = len(genes)
-    adj: List[List[int]] = [[] for _ in range(n)]
-    for s, d in edges:
-        si, di = g2i.get(s), g2i.get(d)
-        if si is None or di is None or si == di:
-            continue
-        adj[si].append(di)
-        adj[di].append(si)
-    rng = np.random.default_rng(seed)
-    module = np.full(n, -1, dtype=np.int64)
-    seeds = rng.choice(n, size=min(n_modules, n), replace=False)
-    queues = [deque([int(s)]) for s in seeds]
-    for k, s in enumerate(seeds):
-        module[int(s)] = k
-    alive = True
-    while alive:
-        alive = False
-        for k, q in enumerate(queues):
-            steps = 0
-            while q and steps < 8:      # round-robin growth keeps sizes even
-                u = q.popleft()
-                for v in adj[u]:
-                    if module[v] < 0:
-                        module[v] = k
-                        q.append(v)
-                        steps += 1
-                alive = True
-    unassigned = module < 0
-    module[unassigned] = np.arange(int(unassigned.sum())) % max(n_modules, 1)
-    if dead_frac > 0:
-        dead = rng.choice(n, size=int(n * dead_frac), replace=False)
-        module[dead] = -1
-    return module
 
 
 def module_activity(n_modules: int, shared_frac: float = 0.2) -> np.ndarray:
