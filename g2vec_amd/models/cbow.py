@@ -198,8 +198,14 @@ class CbowTrainer:
             if on_gpu:
                 st.lrt_buf.fill_(ops.tf1_lr_t(cfg.lr, self.B1, self.B2,
                                               st.t_adam))
-            if (on_gpu and cfg.use_hipgraph and st.graph is None and
-                    not st.graph_failed and st.epoch_idx >= 1):
+            # capture only at world==1 for now: a replay-time RCCL issue
+            # inside a graph could hang rather than raise, and the eager
+            # fallback cannot catch that (collectives are latency-bound at
+            # this message size anyway). TODO round 2: validate RCCL+graphs
+            # on a real multi-GPU node and lift the gate.
+            if (on_gpu and cfg.use_hipgraph and self.ctx.world == 1 and
+                    st.graph is None and not st.graph_failed and
+                    st.epoch_idx >= 1):
                 try:
                     g = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(g):

@@ -98,3 +98,27 @@ def test_missing_file_raises(tmp_path):
                       result_name=str(tmp_path / "out"), device="cpu")
     with pytest.raises(Exception):
         run(cfg)
+
+
+def test_console_transcript_format(tiny_files, tmp_path, capsys):
+    """The phase-numbered console lines ARE the reference's published
+    baseline format (README.md:22-49) — keep them shaped identically."""
+    import re
+    run(_cfg(tiny_files, tmp_path, epochs=6))
+    out = capsys.readouterr().out
+    for pat in (r">>> 1\. Load data",
+                r">>> 2\. Preprocess data",
+                r"    n_samples: \d+",
+                r"    n_genes  : \d+\t\(common genes in both EXPRESSION and NETWORK\)",
+                r"    n_edges  : \d+\t\(edges with the common genes\)",
+                r">>> 3\. Generate random paths from each group",
+                r"    \*\*\* most time consuming step \*\*\*",
+                r"    n_paths : \d+",
+                r">>> 4\. Compute distributed representations using modified CBOW",
+                r"     Start training the modified CBOW with early stopping",
+                r"    - Epoch: 000\tACC\[val\]=0\.\d{4}\tACC\[tr\]=0\.\d{4} \(\d+\.\d{3} sec\)",
+                r"    Optimization Finish",
+                r">>> 5\. Find L-groups",
+                r">>> 6\. Select biomarkers with gene scores",
+                r">>> 7\. Save results"):
+        assert re.search(pat, out), pat
