@@ -90,6 +90,14 @@ class CbowTrainer:
         if self.cfg.seed is not None:
             gen.manual_seed(int(self.cfg.seed) + 12345 +
                             (self.ctx.rank if pre_sharded else 0))
+        elif self.ctx.world > 1 and not pre_sharded:
+            # unseeded DP: every rank must draw the SAME global shuffle or
+            # the train/val shards would overlap/miss paths — agree on a
+            # random seed via broadcast
+            shared = torch.randint(0, 2**31 - 1, (1,),
+                                   device=self.device, dtype=torch.int64)
+            self.ctx.broadcast_(shared)
+            gen.manual_seed(int(shared.item()))
         perm = torch.randperm(P, generator=gen).to(self.device)
         pivot = int(P * 0.8)
         tr_idx, vl_idx = perm[:pivot], perm[pivot:]
