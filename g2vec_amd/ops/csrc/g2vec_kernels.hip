@@ -354,8 +354,18 @@ cbow_fwd_scalar_kernel(const float* __restrict__ s, const int* __restrict__ gene
   for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x; p < P;
        p += (long long)gridDim.x * blockDim.x) {
     const int lo = offs[p], hi = offs[p + 1];
-    float o = 0.f;
-    for (int i = lo; i < hi; ++i) o += s[genes[i]];
+    // 4 independent accumulators keep 4 gathers in flight per lane
+    // (a single running sum serializes on each load's latency)
+    float o0 = 0.f, o1 = 0.f, o2 = 0.f, o3 = 0.f;
+    int i = lo;
+    for (; i + 4 <= hi; i += 4) {
+      o0 += s[genes[i]];
+      o1 += s[genes[i + 1]];
+      o2 += s[genes[i + 2]];
+      o3 += s[genes[i + 3]];
+    }
+    for (; i < hi; ++i) o0 += s[genes[i]];
+    const float o = (o0 + o1) + (o2 + o3);
     const float y = labels[p];
     loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
     correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
@@ -375,8 +385,16 @@ cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ gen
   for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x; p < P;
        p += (long long)gridDim.x * blockDim.x) {
     const int lo = offs[p], hi = offs[p + 1];
-    float o = 0.f;
-    for (int i = lo; i < hi; ++i) o += s[genes[i]];
+    float o0 = 0.f, o1 = 0.f, o2 = 0.f, o3 = 0.f;
+    int i = lo;
+    for (; i + 4 <= hi; i += 4) {
+      o0 += s[genes[i]];
+      o1 += s[genes[i + 1]];
+      o2 += s[genes[i + 2]];
+      o3 += s[genes[i + 3]];
+    }
+    for (; i < hi; ++i) o0 += s[genes[i]];
+    const float o = (o0 + o1) + (o2 + o3);
     const float corr = (((o > 0.f ? 1.f : 0.f) == labels[p]) ? 1.f : 0.f);
     if (p < p_split) c0 += corr; else c1 += corr;
   }
@@ -407,9 +425,16 @@ scatter_do_det_kernel(const int* __restrict__ inst_path,
   for (long long sidx = (long long)blockIdx.x * wpb + wib; sidx < n_seg;
        sidx += (long long)gridDim.x * wpb) {
     const int lo = seg_start[sidx], hi = seg_start[sidx + 1];
-    float partial = 0.f;
-    for (int i = lo + lane; i < hi; i += WAVE) partial += dO[inst_path[i]];
-    const float v = wave_sum(partial);
+    float p0 = 0.f, p1 = 0.f, p2 = 0.f, p3 = 0.f;
+    int i = lo + lane;
+    for (; i + 3 * WAVE < hi; i += 4 * WAVE) {
+      p0 += dO[inst_path[i]];
+      p1 += dO[inst_path[i + WAVE]];
+      p2 += dO[inst_path[i + 2 * WAVE]];
+      p3 += dO[inst_path[i + 3 * WAVE]];
+    }
+    for (; i < hi; i += WAVE) p0 += dO[inst_path[i]];
+    const float v = wave_sum((p0 + p1) + (p2 + p3));
     if (lane == 0) c[seg_gene[sidx]] = v;
   }
 }
