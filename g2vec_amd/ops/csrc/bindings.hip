@@ -86,7 +86,7 @@ std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
   auto correct = torch::empty({P}, opts);
   auto dO = want_grad ? torch::empty({P}, opts) : torch::empty({0}, opts);
   if (P == 0) return {loss, correct, dO};
-  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid_for(P, 256)), dim3(256),
+  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid_for(P, 4)), dim3(256),
                      0, cur_stream(), s.data_ptr<float>(), genes.data_ptr<int>(),
                      offs.data_ptr<int>(), labels.data_ptr<float>(), P,
                      (float)inv_b, loss.data_ptr<float>(),
@@ -107,7 +107,7 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   TORCH_CHECK(counts.numel() == 2, "counts must have 2 elements");
   const long long P = labels.numel();
   if (P == 0) return;
-  hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid_for(P, 256)),
+  hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid_for(P, 4)),
                      dim3(256), 0, cur_stream(), s.data_ptr<float>(),
                      genes.data_ptr<int>(), offs.data_ptr<int>(),
                      labels.data_ptr<float>(), P, (long long)p_split,

@@ -343,66 +343,57 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
 
 // ===================================================== fast path: scalar CBOW
 // o_p = sum_{g in p} s_g; loss = sigmoid-CE(o, y); correct = (o>0)==y;
-// dO = (sigmoid(o)-y)*inv_b. One THREAD per path (paths average ~20 genes;
-// s is L1/L2-resident — the wave-per-path version spent its time on
-// launch ramp and shuffles for ~1 us of work per wave).
+// dO = (sigmoid(o)-y)*inv_b. One WAVE per path: all <= lenPath gathers of a
+// path issue in parallel across lanes (a thread-per-path variant serialized
+// on gather latency and measured 1.5x SLOWER at the 1M-gene scale where the
+// s vector lives in L2/L3).
 extern "C" __global__ void __launch_bounds__(256)
 cbow_fwd_scalar_kernel(const float* __restrict__ s, const int* __restrict__ genes,
                        const int* __restrict__ offs, const float* __restrict__ labels,
                        long long P, float inv_b, float* __restrict__ loss,
                        float* __restrict__ correct, float* __restrict__ dO) {
-  for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x; p < P;
-       p += (long long)gridDim.x * blockDim.x) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
+       p += (long long)gridDim.x * wpb) {
     const int lo = offs[p], hi = offs[p + 1];
-    // 4 independent accumulators keep 4 gathers in flight per lane
-    // (a single running sum serializes on each load's latency)
-    float o0 = 0.f, o1 = 0.f, o2 = 0.f, o3 = 0.f;
-    int i = lo;
-    for (; i + 4 <= hi; i += 4) {
-      o0 += s[genes[i]];
-      o1 += s[genes[i + 1]];
-      o2 += s[genes[i + 2]];
-      o3 += s[genes[i + 3]];
+    float partial = 0.f;
+    for (int i = lo + lane; i < hi; i += WAVE) partial += s[genes[i]];
+    const float o = wave_sum(partial);
+    if (lane == 0) {
+      const float y = labels[p];
+      loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
+      correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
+      if (dO) dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
     }
-    for (; i < hi; ++i) o0 += s[genes[i]];
-    const float o = (o0 + o1) + (o2 + o3);
-    const float y = labels[p];
-    loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
-    correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
-    if (dO) dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
   }
 }
 
 // eval-only variant: nothing stored per path; the two splits' correct
 // counts (train = paths < p_split of the concatenated set, val = rest)
-// reduce in-block and land with two atomicAdds per block
+// accumulate per wave, reduce in-block, and land with two atomicAdds
 extern "C" __global__ void __launch_bounds__(256)
 cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ genes,
                         const int* __restrict__ offs, const float* __restrict__ labels,
                         long long P, long long p_split,
                         float* __restrict__ counts) {
-  float c0 = 0.f, c1 = 0.f;
-  for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x; p < P;
-       p += (long long)gridDim.x * blockDim.x) {
-    const int lo = offs[p], hi = offs[p + 1];
-    float o0 = 0.f, o1 = 0.f, o2 = 0.f, o3 = 0.f;
-    int i = lo;
-    for (; i + 4 <= hi; i += 4) {
-      o0 += s[genes[i]];
-      o1 += s[genes[i + 1]];
-      o2 += s[genes[i + 2]];
-      o3 += s[genes[i + 3]];
-    }
-    for (; i < hi; ++i) o0 += s[genes[i]];
-    const float o = (o0 + o1) + (o2 + o3);
-    const float corr = (((o > 0.f ? 1.f : 0.f) == labels[p]) ? 1.f : 0.f);
-    if (p < p_split) c0 += corr; else c1 += corr;
-  }
-  c0 = wave_sum(c0);
-  c1 = wave_sum(c1);
-  __shared__ float sm[8];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  float c0 = 0.f, c1 = 0.f;                    // lane-0 accumulators
+  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
+       p += (long long)gridDim.x * wpb) {
+    const int lo = offs[p], hi = offs[p + 1];
+    float partial = 0.f;
+    for (int i = lo + lane; i < hi; i += WAVE) partial += s[genes[i]];
+    const float o = wave_sum(partial);
+    if (lane == 0) {
+      const float corr = (((o > 0.f ? 1.f : 0.f) == labels[p]) ? 1.f : 0.f);
+      if (p < p_split) c0 += corr; else c1 += corr;
+    }
+  }
+  __shared__ float sm[8];
   if (lane == 0) { sm[wib] = c0; sm[4 + wib] = c1; }
   __syncthreads();
   if (threadIdx.x == 0) {
