@@ -107,7 +107,12 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   TORCH_CHECK(counts.numel() == 2, "counts must have 2 elements");
   const long long P = labels.numel();
   if (P == 0) return;
-  hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid_for(P, 4)),
+  // cap the grid: every block fan-ins two atomicAdds on the SAME two words
+  // (~11 ns serialized each), and 2048 blocks already fill all 256 CUs —
+  // an uncapped 10k-block launch spent ~200 us in the atomic tail
+  int grid = grid_for(P, 4);
+  if (grid > 2048) grid = 2048;
+  hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid),
                      dim3(256), 0, cur_stream(), s.data_ptr<float>(),
                      genes.data_ptr<int>(), offs.data_ptr<int>(),
                      labels.data_ptr<float>(), P, (long long)p_split,
