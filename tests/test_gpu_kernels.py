@@ -204,3 +204,14 @@ def test_corr_gemm_matches_torch_mm():
 def test_bf16_copy():
     x = torch.randn(1000, device=DEV)
     assert torch.equal(ops.native().bf16_copy(x), x.bfloat16())
+
+
+def test_cbow_eval_counts_matches_oracle():
+    genes, offs, labels = _pathset_tensors(seed=8)
+    s = torch.randn(300)
+    counts = torch.zeros(2, device=DEV)
+    ops.cbow_eval_counts_(s.to(DEV), genes.to(DEV), offs.to(DEV),
+                          labels.to(DEV), 300, counts)
+    _l, corr, _d = cpu_ref.cbow_fwd_scalar(s, genes, offs, labels, 1.0, False)
+    assert float(counts[0]) == pytest.approx(float(corr[:300].sum()))
+    assert float(counts[1]) == pytest.approx(float(corr[300:].sum()))
