@@ -108,11 +108,12 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   const long long P = labels.numel();
   if (P == 0) return;
   // cap the grid: every block fan-ins two atomicAdds on the SAME two words
-  // (~20 ns serialized each); 256 blocks = 1024 waves amply cover the path
-  // set via the grid-stride loop —
-  // an uncapped 10k-block launch spent ~200 us in the atomic tail
+  // (~11 ns serialized each; an uncapped 10k-block launch spent ~200 us in
+  // the atomic tail). Small path sets: 256 blocks keep the tail ~6 us;
+  // large sets: 2048 blocks fill the chip and amortize the ~45 us tail.
   int grid = grid_for(P, 4);
-  if (grid > 256) grid = 256;
+  const int cap = (P > 65536) ? 2048 : 256;
+  if (grid > cap) grid = cap;
   hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid),
                      dim3(256), 0, cur_stream(), s.data_ptr<float>(),
                      genes.data_ptr<int>(), offs.data_ptr<int>(),
