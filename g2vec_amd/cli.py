@@ -37,6 +37,10 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--dtype", choices=["fp32", "bf16", "fp16"], default="bf16")
     p.add_argument("--device", choices=["auto", "cpu", "cuda"], default="auto")
     p.add_argument("--pcc-mode", choices=["auto", "edge", "gemm"], default="auto")
+    p.add_argument("--pcc-threshold", type=float, default=0.5,
+                   help="|PCC| cutoff for graph edges (reference: 0.5)")
+    p.add_argument("--kmeans", choices=["auto", "sklearn", "torch"],
+                   default="auto", help="L-group clustering backend")
     p.add_argument("--trainer-path", choices=["fast", "general"], default="fast")
     p.add_argument("--batch-size", type=int, default=0,
                    help="0 = full batch (reference semantics)")
@@ -62,7 +66,8 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         hidden=a.sizeHiddenlayer, epochs=a.epoch, lr=a.learningRate,
         num_biomarker=a.numBiomarker,
         seed=(None if a.seed < 0 else a.seed), dtype=a.dtype, device=a.device,
-        pcc_mode=a.pcc_mode, trainer_path=a.trainer_path,
+        pcc_mode=a.pcc_mode, pcc_threshold=a.pcc_threshold,
+        kmeans_backend=a.kmeans, trainer_path=a.trainer_path,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, save_paths=a.save_paths,
         load_paths=a.load_paths, save_model=a.save_model,

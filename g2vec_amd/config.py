@@ -30,6 +30,7 @@ class G2VecConfig:
     device: str = "auto"            # "auto" | "cpu" | "cuda"
     pcc_threshold: float = 0.5      # |PCC| cutoff (G2Vec.py:385-390)
     pcc_mode: str = "auto"          # "edge" (per-edge dot) | "gemm" (MFMA corr GEMM) | "auto"
+    kmeans_backend: str = "auto"    # "sklearn" (reference parity) | "torch" (scales) | "auto"
     compat_lgroup_bug: bool = False  # reproduce the shipped G2Vec.py:186-194 behaviour
     early_stop: bool = True
     batch_size: int = 0             # 0 = full batch (reference semantics, G2Vec.py:264)
@@ -55,6 +56,8 @@ class G2VecConfig:
             raise ValueError(f"bad pcc_mode {self.pcc_mode}")
         if self.trainer_path not in ("fast", "general"):
             raise ValueError(f"bad trainer_path {self.trainer_path}")
+        if self.kmeans_backend not in ("auto", "sklearn", "torch"):
+            raise ValueError(f"bad kmeans_backend {self.kmeans_backend}")
 
 
 def resolve_device(device: str) -> str:

@@ -144,7 +144,7 @@ def run(cfg: G2VecConfig, ctx: Optional[DistContext] = None) -> Dict:
     log(">>> 5. Find L-groups")
     with timers.phase("lgroups"):
         lg = find_lgroups(W, freq.cpu().numpy(), cfg.compat_lgroup_bug,
-                          device=device)
+                          backend=cfg.kmeans_backend, device=device)
 
     log(">>> 6. Select biomarkers with gene scores")
     with timers.phase("scoring"):

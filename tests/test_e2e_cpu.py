@@ -122,3 +122,12 @@ def test_console_transcript_format(tiny_files, tmp_path, capsys):
                 r">>> 6\. Select biomarkers with gene scores",
                 r">>> 7\. Save results"):
         assert re.search(pat, out), pat
+
+
+def test_save_model_checkpoint(tiny_files, tmp_path):
+    ckpt = str(tmp_path / "model.pt")
+    res = run(_cfg(tiny_files, tmp_path, save_model=ckpt, epochs=5))
+    blob = torch.load(ckpt, weights_only=False)
+    assert blob["W_ih"].shape == (res["n_genes"], 128)
+    assert blob["acc_val"] == res["acc_val"]
+    assert len(blob["gene_index"]) == res["n_genes"]
