@@ -86,7 +86,10 @@ std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
   auto correct = torch::empty({P}, opts);
   auto dO = want_grad ? torch::empty({P}, opts) : torch::empty({0}, opts);
   if (P == 0) return {loss, correct, dO};
-  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid_for(P, 4)), dim3(256),
+  int grid = grid_for(P, 4);
+  if (grid > 2048) grid = 2048;      // dispatch ramp of 8k+ tiny blocks costs
+                                     // more than the grid-stride work itself
+  hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid), dim3(256),
                      0, cur_stream(), s.data_ptr<float>(), genes.data_ptr<int>(),
                      offs.data_ptr<int>(), labels.data_ptr<float>(), P,
                      (float)inv_b, loss.data_ptr<float>(),
@@ -107,17 +110,17 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   TORCH_CHECK(counts.numel() == 2, "counts must have 2 elements");
   const long long P = labels.numel();
   if (P == 0) return;
-  // cap the grid: every block fan-ins two atomicAdds on the SAME two words
-  // (~11 ns serialized each; an uncapped 10k-block launch spent ~200 us in
-  // the atomic tail). Small path sets: 256 blocks keep the tail ~6 us;
-  // large sets: 2048 blocks fill the chip and amortize the ~45 us tail.
   int grid = grid_for(P, 4);
-  const int cap = (P > 65536) ? 2048 : 256;
-  if (grid > cap) grid = cap;
+  if (grid > 2048) grid = 2048;      // 8192 waves fill the chip
+  auto partials = torch::empty({grid, 2},
+      torch::TensorOptions().dtype(at::kFloat).device(s.device()));
   hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid),
                      dim3(256), 0, cur_stream(), s.data_ptr<float>(),
                      genes.data_ptr<int>(), offs.data_ptr<int>(),
                      labels.data_ptr<float>(), P, (long long)p_split,
+                     partials.data_ptr<float>());
+  hipLaunchKernelGGL(fold_partials_kernel, dim3(1), dim3(256), 0,
+                     cur_stream(), partials.data_ptr<float>(), grid,
                      counts.data_ptr<float>());
   LAUNCH_CHECK();
 }
