@@ -86,7 +86,7 @@ std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
   auto correct = torch::empty({P}, opts);
   auto dO = want_grad ? torch::empty({P}, opts) : torch::empty({0}, opts);
   if (P == 0) return {loss, correct, dO};
-  int grid = grid_for(P, 4);
+  int grid = grid_for(P, 16);        // 16 paths per block (16-lane sub-waves)
   if (grid > 2048) grid = 2048;      // dispatch ramp of 8k+ tiny blocks costs
                                      // more than the grid-stride work itself
   hipLaunchKernelGGL(cbow_fwd_scalar_kernel, dim3(grid), dim3(256),
@@ -110,7 +110,7 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   TORCH_CHECK(counts.numel() == 2, "counts must have 2 elements");
   const long long P = labels.numel();
   if (P == 0) return;
-  int grid = grid_for(P, 4);
+  int grid = grid_for(P, 16);       // 16 paths per block (16-lane sub-waves)
   if (grid > 2048) grid = 2048;      // 8192 waves fill the chip
   auto partials = torch::empty({grid, 2},
       torch::TensorOptions().dtype(at::kFloat).device(s.device()));

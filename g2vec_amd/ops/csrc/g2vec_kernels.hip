@@ -343,25 +343,35 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
 
 // ===================================================== fast path: scalar CBOW
 // o_p = sum_{g in p} s_g; loss = sigmoid-CE(o, y); correct = (o>0)==y;
-// dO = (sigmoid(o)-y)*inv_b. One WAVE per path: all <= lenPath gathers of a
-// path issue in parallel across lanes (a thread-per-path variant serialized
-// on gather latency and measured 1.5x SLOWER at the 1M-gene scale where the
-// s vector lives in L2/L3).
+// dO = (sigmoid(o)-y)*inv_b. One 16-LANE SUB-WAVE per path (4 paths per
+// wavefront): paths average ~21 genes, so a full 64-lane wave left 2/3 of
+// its lanes idle in both the gather and the reduce, and a thread-per-path
+// variant serialized on gather latency (1.5x slower at 1M genes).
+#define SUBW 16
+
+__device__ __forceinline__ float subwave_sum16(float v) {
+  v += __shfl_xor(v, 8);
+  v += __shfl_xor(v, 4);
+  v += __shfl_xor(v, 2);
+  v += __shfl_xor(v, 1);
+  return v;                       // every lane of the 16-group has the sum
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 cbow_fwd_scalar_kernel(const float* __restrict__ s, const int* __restrict__ genes,
                        const int* __restrict__ offs, const float* __restrict__ labels,
                        long long P, float inv_b, float* __restrict__ loss,
                        float* __restrict__ correct, float* __restrict__ dO) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wib = threadIdx.x >> 6;
-  const int wpb = blockDim.x >> 6;
-  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
-       p += (long long)gridDim.x * wpb) {
+  const int sublane = threadIdx.x & (SUBW - 1);
+  const int subs_per_block = blockDim.x / SUBW;
+  const int sub = threadIdx.x / SUBW;
+  for (long long p = (long long)blockIdx.x * subs_per_block + sub; p < P;
+       p += (long long)gridDim.x * subs_per_block) {
     const int lo = offs[p], hi = offs[p + 1];
     float partial = 0.f;
-    for (int i = lo + lane; i < hi; i += WAVE) partial += s[genes[i]];
-    const float o = wave_sum(partial);
-    if (lane == 0) {
+    for (int i = lo + sublane; i < hi; i += SUBW) partial += s[genes[i]];
+    const float o = subwave_sum16(partial);
+    if (sublane == 0) {
       const float y = labels[p];
       loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
       correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
@@ -380,22 +390,28 @@ cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ gen
                         const int* __restrict__ offs, const float* __restrict__ labels,
                         long long P, long long p_split,
                         float* __restrict__ partials) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wib = threadIdx.x >> 6;
-  const int wpb = blockDim.x >> 6;
-  float c0 = 0.f, c1 = 0.f;                    // lane-0 accumulators
-  for (long long p = (long long)blockIdx.x * wpb + wib; p < P;
-       p += (long long)gridDim.x * wpb) {
+  const int sublane = threadIdx.x & (SUBW - 1);
+  const int subs_per_block = blockDim.x / SUBW;
+  const int sub = threadIdx.x / SUBW;
+  float c0 = 0.f, c1 = 0.f;                    // sublane-0 accumulators
+  for (long long p = (long long)blockIdx.x * subs_per_block + sub; p < P;
+       p += (long long)gridDim.x * subs_per_block) {
     const int lo = offs[p], hi = offs[p + 1];
     float partial = 0.f;
-    for (int i = lo + lane; i < hi; i += WAVE) partial += s[genes[i]];
-    const float o = wave_sum(partial);
-    if (lane == 0) {
+    for (int i = lo + sublane; i < hi; i += SUBW) partial += s[genes[i]];
+    const float o = subwave_sum16(partial);
+    if (sublane == 0) {
       const float corr = (((o > 0.f ? 1.f : 0.f) == labels[p]) ? 1.f : 0.f);
       if (p < p_split) c0 += corr; else c1 += corr;
     }
   }
+  // fold the per-thread accumulators: sublane0 lanes hold the values;
+  // wave-reduce then block-reduce
+  c0 = wave_sum(c0);
+  c1 = wave_sum(c1);
   __shared__ float sm[8];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
   if (lane == 0) { sm[wib] = c0; sm[4 + wib] = c1; }
   __syncthreads();
   if (threadIdx.x == 0) {
