@@ -148,7 +148,12 @@ class CbowTrainer:
         st.t_adam = 0
         st.epoch_idx = 0
         # persistent fast-path buffers (stable addresses across hipGraph replays)
-        st.s_buf = torch.mv(W, who) if not use_general else None
+        st.s_buf = None
+        st.gradwho_buf = torch.empty_like(who)
+        if not use_general:
+            st.s_buf = torch.empty(self.G, dtype=torch.float32,
+                                   device=self.device)
+            ops.gemv_rows(W, who, st.s_buf)
         st.lrt_buf = torch.zeros(1, dtype=torch.float32, device=self.device)
         st.counts_buf = torch.zeros(2, dtype=torch.float32, device=self.device)
         st.graph = None
@@ -177,14 +182,14 @@ class CbowTrainer:
             st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True)
         c = ops.scatter_dO(tr.genes, tr.offsets, dO, self.G, plan=st.plan)
         self.ctx.allreduce_(c)                  # C1: whole dW_ih message
-        grad_who = torch.mv(st.W.t(), c)        # dW_ho = W_ih^T c (pre-update W)
+        ops.gemv_cols(st.W, c, st.gradwho_buf)  # dW_ho = W_ih^T c (pre-update W)
         ops.adam_rank1(st.W, st.mW, st.vW, c, st.who, st.t_adam, cfg.lr,
                        self.B1, self.B2, self.EPS, lrt_buf=lrt)
-        ops.adam_dense(st.who, st.mO, st.vO, grad_who, st.t_adam, cfg.lr,
-                       self.B1, self.B2, self.EPS, lrt_buf=lrt)
+        ops.adam_dense(st.who, st.mO, st.vO, st.gradwho_buf, st.t_adam,
+                       cfg.lr, self.B1, self.B2, self.EPS, lrt_buf=lrt)
         # post-update accuracy (reference order, G2Vec.py:264-267):
         # one fused eval kernel over the concatenated train+val paths
-        torch.mv(st.W, st.who, out=st.s_buf)
+        ops.gemv_rows(st.W, st.who, st.s_buf)
         if st.ev_genes is not None:
             ops.cbow_eval_counts_(st.s_buf, st.ev_genes, st.ev_offsets,
                                   st.ev_labels, tr.n_paths, st.counts_buf)
@@ -245,7 +250,7 @@ class CbowTrainer:
             acc_tr = self._accuracy(st.W, st.W16, st.who, st.tr, self.n_tr_global)
             acc_val = self._accuracy(st.W, st.W16, st.who, st.vl, self.n_vl_global)
         else:
-            torch.mv(st.W, st.who, out=st.s_buf)    # post-update s
+            ops.gemv_rows(st.W, st.who, st.s_buf)   # post-update s
             counts = torch.empty(2, dtype=torch.float32, device=self.device)
             for k, split in enumerate((st.tr, st.vl)):
                 if split.n_paths == 0:
@@ -322,17 +327,17 @@ class CbowTrainer:
         W, who, tr = st.W, st.who, st.tr
         genes, offsets, labels = self._slice(tr, lo, hi)
         if lo != 0:                 # s_buf is fresh only for the first batch
-            torch.mv(W, who, out=st.s_buf)
+            ops.gemv_rows(W, who, st.s_buf)
         _loss, _corr, dO = ops.cbow_fwd_scalar(st.s_buf, genes, offsets,
                                                labels, inv_b, True)
         use_plan = st.plan if (lo == 0 and hi == tr.n_paths) else None
         c = ops.scatter_dO(genes, offsets, dO, self.G, plan=use_plan)
         self.ctx.allreduce_(c)                      # C1: the whole dW_ih message
-        grad_who = torch.mv(W.t(), c)               # dW_ho = W_ih^T c (pre-update W)
+        ops.gemv_cols(W, c, st.gradwho_buf)         # dW_ho = W_ih^T c (pre-update W)
         ops.adam_rank1(W, st.mW, st.vW, c, who, t, self.cfg.lr, self.B1,
                        self.B2, self.EPS)
-        ops.adam_dense(who, st.mO, st.vO, grad_who, t, self.cfg.lr, self.B1,
-                       self.B2, self.EPS)
+        ops.adam_dense(who, st.mO, st.vO, st.gradwho_buf, t, self.cfg.lr,
+                       self.B1, self.B2, self.EPS)
 
     def _step_general(self, st, lo, hi, inv_b, t):
         W, W16, who, tr = st.W, st.W16, st.who, st.tr

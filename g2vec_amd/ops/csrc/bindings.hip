@@ -316,6 +316,64 @@ torch::Tensor corr_gemm(torch::Tensor zt, int64_t n_group) {
   return C;
 }
 
+void gemv_rows_(torch::Tensor W, torch::Tensor x, torch::Tensor out) {
+  CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
+  CHECK_DEV(x); CHECK_CONT(x); CHECK_F32(x);
+  CHECK_DEV(out); CHECK_CONT(out); CHECK_F32(out);
+  const long long G = W.size(0);
+  const int h = (int)W.size(1);
+  const int hpl = h / 64;
+  TORCH_CHECK(h % 64 == 0 && hpl >= 1 && hpl <= 16 && (hpl & (hpl - 1)) == 0,
+              "hidden must be 64*{1,2,4,8,16}");
+  TORCH_CHECK(out.numel() == G && x.numel() == h, "gemv_rows shape mismatch");
+  int grid = grid_for(G, 4);
+  if (grid > 8192) grid = 8192;
+#define GEMVR_CASE(HPL)                                                       \
+  hipLaunchKernelGGL((gemv_rows_kernel<HPL>), dim3(grid), dim3(256), 0,       \
+                     cur_stream(), W.data_ptr<float>(), x.data_ptr<float>(),  \
+                     G, h, out.data_ptr<float>())
+  switch (hpl) {
+    case 1: GEMVR_CASE(1); break;
+    case 2: GEMVR_CASE(2); break;
+    case 4: GEMVR_CASE(4); break;
+    case 8: GEMVR_CASE(8); break;
+    default: GEMVR_CASE(16); break;
+  }
+#undef GEMVR_CASE
+  LAUNCH_CHECK();
+}
+
+void gemv_cols_(torch::Tensor W, torch::Tensor c, torch::Tensor out) {
+  CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
+  CHECK_DEV(c); CHECK_CONT(c); CHECK_F32(c);
+  CHECK_DEV(out); CHECK_CONT(out); CHECK_F32(out);
+  const long long G = W.size(0);
+  const int h = (int)W.size(1);
+  const int cpt = (h >= 256) ? h / 256 : 1;
+  TORCH_CHECK(h % 64 == 0 && cpt >= 1 && cpt <= 4 && (cpt & (cpt - 1)) == 0,
+              "gemv_cols needs hidden 64*k with h/256 in {<=1,2,4}");
+  TORCH_CHECK(out.numel() == h && c.numel() == G, "gemv_cols shape mismatch");
+  int grid = (int)((G + 1023) / 1024);
+  if (grid > 1024) grid = 1024;
+  if (grid < 1) grid = 1;
+  auto partials = torch::empty({grid, h},
+      torch::TensorOptions().dtype(at::kFloat).device(W.device()));
+#define GEMVC_CASE(CPT)                                                       \
+  hipLaunchKernelGGL((gemv_cols_kernel<CPT>), dim3(grid), dim3(256), 0,       \
+                     cur_stream(), W.data_ptr<float>(), c.data_ptr<float>(),  \
+                     G, h, partials.data_ptr<float>())
+  switch (cpt) {
+    case 1: GEMVC_CASE(1); break;
+    case 2: GEMVC_CASE(2); break;
+    default: GEMVC_CASE(4); break;
+  }
+#undef GEMVC_CASE
+  hipLaunchKernelGGL(fold_cols_kernel, dim3((h + 255) / 256), dim3(256), 0,
+                     cur_stream(), partials.data_ptr<float>(), grid, h,
+                     out.data_ptr<float>());
+  LAUNCH_CHECK();
+}
+
 torch::Tensor bf16_copy(torch::Tensor src) {
   CHECK_DEV(src); CHECK_CONT(src); CHECK_F32(src);
   auto out = torch::empty_like(src, src.options().dtype(at::kBFloat16));
@@ -391,6 +449,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cbow_bwd_rows", &cbow_bwd_rows, "scatter-add CBOW backward");
   m.def("pcc_edges", &pcc_edges, "per-edge |PCC|");
   m.def("corr_gemm", &corr_gemm, "MFMA f32 correlation GEMM");
+  m.def("gemv_rows_", &gemv_rows_, "s = W @ x (wave-per-row GEMV, out-arg)");
+  m.def("gemv_cols_", &gemv_cols_, "out = W^T c (partials + fold, out-arg)");
   m.def("bf16_copy", &bf16_copy, "f32 -> bf16 cast kernel");
   m.def("parse_expression_tsv", &parse_expression_tsv,
         "native expression TSV parser");

@@ -215,3 +215,17 @@ def test_cbow_eval_counts_matches_oracle():
     _l, corr, _d = cpu_ref.cbow_fwd_scalar(s, genes, offs, labels, 1.0, False)
     assert float(counts[0]) == pytest.approx(float(corr[:300].sum()))
     assert float(counts[1]) == pytest.approx(float(corr[300:].sum()))
+
+
+@pytest.mark.parametrize("G,h", [(1000, 64), (777, 128), (513, 256), (300, 512)])
+def test_gemv_kernels_match_torch(G, h):
+    torch.manual_seed(3)
+    W = torch.randn(G, h, device=DEV)
+    x = torch.randn(h, device=DEV)
+    c = torch.randn(G, device=DEV)
+    s = torch.empty(G, device=DEV)
+    ops.gemv_rows(W, x, s)
+    assert torch.allclose(s, torch.mv(W, x), atol=1e-4)
+    g = torch.empty(h, device=DEV)
+    ops.gemv_cols(W, c, g)
+    assert torch.allclose(g, torch.mv(W.t(), c), atol=1e-3)
