@@ -370,7 +370,7 @@ void gemv_cols_(torch::Tensor W, torch::Tensor c, torch::Tensor out) {
     default: GEMVC_CASE(4); break;
   }
 #undef GEMVC_CASE
-  hipLaunchKernelGGL(fold_cols_kernel, dim3((h + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(fold_cols_kernel, dim3((h + 3) / 4), dim3(256), 0,
                      cur_stream(), partials.data_ptr<float>(), grid, h,
                      out.data_ptr<float>());
   LAUNCH_CHECK();

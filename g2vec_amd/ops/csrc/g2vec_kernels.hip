@@ -753,12 +753,16 @@ gemv_cols_kernel(const float* __restrict__ W, const float* __restrict__ c,
 extern "C" __global__ void __launch_bounds__(256)
 fold_cols_kernel(const float* __restrict__ partials, int n_blocks, int h,
                  float* __restrict__ out) {
-  for (int j = blockIdx.x * blockDim.x + threadIdx.x; j < h;
-       j += gridDim.x * blockDim.x) {
-    float a = 0.f;
-    for (int b = 0; b < n_blocks; ++b) a += partials[(long long)b * h + j];
-    out[j] = a;
-  }
+  // one wave per column, lane-strided over the block partials (a serial
+  // per-thread loop over ~1k uncoalesced partial rows measured ~150 us)
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int col = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (col >= h) return;
+  float a = 0.f;
+  for (int b = lane; b < n_blocks; b += WAVE)
+    a += partials[(long long)b * h + col];
+  a = wave_sum(a);
+  if (lane == 0) out[col] = a;
 }
 
 // =============================================================== bf16 cast
