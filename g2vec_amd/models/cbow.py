@@ -114,6 +114,14 @@ class CbowTrainer:
             vl_idx = vl_idx[self.ctx.shard_indices(vl_idx.numel(), self.device)]
         return subset(ps, tr_idx), subset(ps, vl_idx)
 
+    @staticmethod
+    def _locality_sort(ps: PathSet) -> PathSet:
+        if ps.n_paths == 0:
+            return ps
+        first = ps.genes[ps.offsets[:-1].long()].long()
+        order = torch.argsort(first, stable=True)
+        return subset(ps, order)
+
     # ------------------------------------------------------------------ setup
     def setup(self, ps: PathSet, pre_sharded: bool = False):
         """Initialise weights/optimizer state and split the path set.
@@ -129,6 +137,13 @@ class CbowTrainer:
         tr, vl = self._split(ps, pre_sharded)
 
         use_general = cfg.trainer_path == "general"
+        if cfg.batch_size == 0:
+            # full batch: within-shard path ORDER is free (all math is a sum
+            # over paths) — sort paths by first gene so co-module paths are
+            # processed by adjacent sub-waves and the s/dO gathers hit
+            # cache-resident slices instead of the whole table
+            tr = self._locality_sort(tr)
+            vl = self._locality_sort(vl)
         st = type("TrainState", (), {})()
         st.W, st.who = W, who
         st.mW, st.vW = torch.zeros_like(W), torch.zeros_like(W)
