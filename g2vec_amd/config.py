@@ -44,10 +44,10 @@ class G2VecConfig:
     use_hipgraph: bool = True       # record the full-batch epoch into a hipGraph
 
     def validate(self) -> None:
-        if self.hidden % 64 != 0 or not (64 <= self.hidden <= 1024):
+        if self.hidden not in (64, 128, 256, 512, 1024):
             raise ValueError(
-                f"hidden={self.hidden}: MI355X kernels require a multiple of the "
-                f"64-lane wavefront in [64, 1024]")
+                f"hidden={self.hidden}: MI355X kernels require a power-of-two "
+                f"multiple of the 64-lane wavefront in {{64,128,256,512,1024}}")
         if self.dtype not in ("fp32", "bf16", "fp16"):
             raise ValueError(f"dtype must be fp32|bf16|fp16, got {self.dtype}")
         if self.len_path < 1 or self.len_path > 512:
