@@ -262,22 +262,28 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
           int chosen_cand = -1;
           float base = 0.f;
           int last_pos_cand = -1;
-          for (int k = 0; k < nchunk; ++k) {
-            const float w = wc[t][k];
-            const float scan = wave_incl_scan(w);
-            const float chunk_tot = unif(__shfl(scan, WAVE - 1));
-            const bool hit = (w > 0.f) && (base + scan > target) &&
-                             (base + scan - w <= target);
-            const unsigned long long m = __ballot(hit);
-            if (m != 0ULL) {
-              chosen_cand = uni(__shfl(cand[t][k], __ffsll((long long)m) - 1));
-              break;
+          // fully unrolled with uniform guards: a runtime-indexed chunk
+          // loop sends cand/wc to scratch (rule-20 trap; measured 3x)
+#pragma unroll
+          for (int k = 0; k < WCHUNKS; ++k) {
+            if (k < nchunk && chosen_cand < 0) {
+              const float w = wc[t][k];
+              const float scan = wave_incl_scan(w);
+              const float chunk_tot = unif(__shfl(scan, WAVE - 1));
+              const bool hit = (w > 0.f) && (base + scan > target) &&
+                               (base + scan - w <= target);
+              const unsigned long long m = __ballot(hit);
+              if (m != 0ULL) {
+                chosen_cand = uni(__shfl(cand[t][k],
+                                         __ffsll((long long)m) - 1));
+              } else {
+                const unsigned long long mp = __ballot(w > 0.f);
+                if (mp != 0ULL)
+                  last_pos_cand = uni(__shfl(cand[t][k],
+                                             63 - __clzll((long long)mp)));
+                base += chunk_tot;
+              }
             }
-            const unsigned long long mp = __ballot(w > 0.f);
-            if (mp != 0ULL)
-              last_pos_cand = uni(__shfl(cand[t][k],
-                                         63 - __clzll((long long)mp)));
-            base += chunk_tot;
           }
           if (chosen_cand < 0) {
             // rounding tail: target >= running total -> last unvisited
