@@ -59,10 +59,8 @@ std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col
   const int wpb = 4;                 // 256 threads = 4 waves
   int tsize = 256;                   // LDS hash set: >= 2x path length, pow2
   while (tsize < 2 * (int)len_path) tsize <<= 1;
-  // two walks per wave (paired phase-interleaved kernel): double LDS
-  const size_t lds = (size_t)wpb * 2 * (len_path + tsize) * sizeof(int);
-  const long long n_pairs = (n_walks + 1) / 2;
-  hipLaunchKernelGGL(walk_kernel, dim3(grid_for(n_pairs, wpb)), dim3(256), lds,
+  const size_t lds = (size_t)wpb * (len_path + tsize) * sizeof(int);
+  hipLaunchKernelGGL(walk_kernel, dim3(grid_for(n_walks, wpb)), dim3(256), lds,
                      cur_stream(), row_ptr.data_ptr<int>(),
                      col_idx.data_ptr<int>(), weights.data_ptr<float>(),
                      sources.data_ptr<int>(),

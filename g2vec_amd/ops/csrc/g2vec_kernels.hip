@@ -163,199 +163,180 @@ walk_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
             int n_src, long long n_walks, int num_rep, int len_path, int tsize,
             uint64_t seed, int* __restrict__ out_nodes,
             int* __restrict__ out_len, long long* __restrict__ out_hash) {
-  // TWO walks per wave, phase-interleaved: the per-step serial chain
-  // (row fetch -> candidate loads -> LDS probe -> scan/select) is
-  // latency-bound (~47% SQ_WAIT_ANY measured), so the two independent
-  // chains issue their loads back-to-back (phases 1-2) before either
-  // consumes (phase 3) and their latencies overlap. All slot/chunk loops
-  // are fully unrolled (constant trips) so slot state stays in registers.
   extern __shared__ int smem[];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wib = uni(threadIdx.x >> 6);       // wave-in-block (SGPR)
   const int wpb = blockDim.x >> 6;             // waves per block
-  int* vis0 = smem + wib * 2 * len_path;       // ordered paths (output)
-  uint32_t* tab0 = (uint32_t*)(smem + wpb * 2 * len_path) + wib * 2 * tsize;
+  int* vis = smem + wib * len_path;            // ordered path (output)
+  uint32_t* tab = (uint32_t*)(smem + wpb * len_path) + wib * tsize;
   const uint32_t tmask = (uint32_t)tsize - 1;
 
-  // per-slot state (constant-indexed only)
-  bool alive[2];
-  int cur[2], plen[2], rs[2], deg[2];
-  uint64_t state[2], hash[2];
-  int cand[2][WCHUNKS];
-  float wc[2][WCHUNKS];
+  for (long long walk = (long long)blockIdx.x * wpb + wib; walk < n_walks;
+       walk += (long long)gridDim.x * wpb) {
+    const int rep = (int)(walk / n_src);
+    const int src = uni(sources[walk % n_src]);
+    // RNG keyed on the GLOBAL (source, repetition): DP-sharded generation
+    // is bitwise-identical to single-process (C5 rank invariance)
+    const uint64_t gid = (uint64_t)src * (uint64_t)num_rep + (uint64_t)rep;
+    uint64_t state = seed ^ (uint64_t)(gid * 0x94D049BB133111EBULL + 1ULL);
+    (void)sm64_next(state);                    // warm draw (CPU oracle parity)
+    int cur = src;
+    int plen = 0;
+    uint64_t hash = 0;
+    for (int i = lane; i < tsize; i += WAVE) tab[i] = HSET_EMPTY;
 
-  const long long stride = (long long)gridDim.x * wpb * 2;
-  for (long long base0 = ((long long)blockIdx.x * wpb + wib) * 2;
-       base0 < n_walks; base0 += stride) {
-#pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      const long long wk = base0 + t;
-      alive[t] = wk < n_walks;
-      plen[t] = 0;
-      hash[t] = 0;
-      if (alive[t]) {
-        const int rep = (int)(wk / n_src);
-        const int src = uni(sources[wk % n_src]);
-        // RNG keyed on the GLOBAL (source, repetition): DP-sharded
-        // generation is bitwise-identical to single-process
-        const uint64_t gid = (uint64_t)src * (uint64_t)num_rep + (uint64_t)rep;
-        state[t] = seed ^ (uint64_t)(gid * 0x94D049BB133111EBULL + 1ULL);
-        (void)sm64_next(state[t]);             // warm draw (CPU oracle parity)
-        cur[t] = src;
-        uint32_t* tab = tab0 + t * tsize;
-        for (int i = lane; i < tsize; i += WAVE) tab[i] = HSET_EMPTY;
+    for (int step = 0; step < len_path; ++step) {
+      G2V_ASSERT(plen < len_path && cur >= 0);
+      if (lane == 0) {
+        vis[plen] = cur;
+        // ONE lane inserts: a 64-lane same-address LDS store serializes its
+        // lane group and the LDS array is shared by all 32 waves of the CU
+        hset_insert(tab, tmask, (uint32_t)cur);
       }
-    }
+      ++plen;
+      hash += gene_hash_dev((uint32_t)cur);
+      const int s = row_ptr[cur], e = row_ptr[cur + 1];
+      const int deg = e - s;
+      if (deg <= 0) break;
 
-    for (int step = 0; (alive[0] || alive[1]) && step < len_path; ++step) {
-      // ---- phase 1: bookkeeping + row fetch (scalar loads) per slot
-#pragma unroll
-      for (int t = 0; t < 2; ++t) {
-        if (!alive[t]) continue;
-        G2V_ASSERT(plen[t] < len_path && cur[t] >= 0);
-        if (lane == 0) {
-          vis0[t * len_path + plen[t]] = cur[t];
-          // ONE lane inserts: a 64-lane same-address LDS store serializes
-          // its lane group and the LDS array is shared CU-wide
-          hset_insert(tab0 + t * tsize, tmask, (uint32_t)cur[t]);
+      if (deg <= WAVE) {
+        // leanest path (the common case): one candidate per lane
+        const int j = lane;
+        int cand = -1;
+        float w = 0.f;
+        if (j < deg) {
+          cand = col_idx[s + j];
+          w = wgt[s + j];
+          if (hset_contains(tab, tmask, (uint32_t)cand)) w = 0.f;
         }
-        ++plen[t];
-        hash[t] += gene_hash_dev((uint32_t)cur[t]);
-        rs[t] = row_ptr[cur[t]];
-        deg[t] = row_ptr[cur[t] + 1] - rs[t];
-        if (deg[t] <= 0) alive[t] = false;     // dead end before any draw
-      }
-      // ---- phase 2: raw candidate loads for both slots (<= 4 chunks)
-#pragma unroll
-      for (int t = 0; t < 2; ++t) {
-        if (!(alive[t] && deg[t] <= WCHUNKS * WAVE)) continue;
+        // one scan serves both the total (its lane-63 element) and the
+        // selection — 6 fewer dependent ds_bpermute per step
+        const float scan = wave_incl_scan(w);
+        const float tot = unif(__shfl(scan, WAVE - 1));
+        const uint64_t r = sm64_next(state);   // drawn even on dead end
+        if (!(tot > 0.f)) break;
+        const float target = (float)(u01_from(r) * (double)tot);
+        const bool hit = (w > 0.f) && (scan > target) && (scan - w <= target);
+        const unsigned long long mh = __ballot(hit);
+        int lane_sel;
+        if (mh != 0ULL) {
+          lane_sel = __ffsll((long long)mh) - 1;
+        } else {
+          const unsigned long long mp = __ballot(w > 0.f);
+          lane_sel = 63 - __clzll((long long)mp);
+        }
+        cur = uni(__shfl(cand, lane_sel));
+      } else if (deg <= WCHUNKS * WAVE) {
+        // register path: the whole row (<= 4 chunks of 64) is loaded once,
+        // membership-masked once, and both the total and the selection use
+        // the cached registers — one dwordx2 load + one hash probe per
+        // candidate per STEP, nothing re-read.
+        const int nchunk = (deg + WAVE - 1) >> 6;
+        float wreg[WCHUNKS];
+        int creg[WCHUNKS];
+        float partial = 0.f;
 #pragma unroll
         for (int k = 0; k < WCHUNKS; ++k) {
+          wreg[k] = 0.f;
+          creg[k] = -1;
           const int j = (k << 6) + lane;
-          cand[t][k] = -1;
-          wc[t][k] = 0.f;
-          if (j < deg[t]) {
-            cand[t][k] = col_idx[rs[t] + j];
-            wc[t][k] = wgt[rs[t] + j];
-          }
-        }
-      }
-      // ---- phase 3: mask + sample per slot (loads above now in flight)
-#pragma unroll
-      for (int t = 0; t < 2; ++t) {
-        if (!alive[t]) continue;
-        const uint32_t* tab = tab0 + t * tsize;
-        if (deg[t] <= WCHUNKS * WAVE) {
-          const int nchunk = (deg[t] + WAVE - 1) >> 6;
-          float partial = 0.f;
-#pragma unroll
-          for (int k = 0; k < WCHUNKS; ++k) {
-            if (cand[t][k] >= 0 &&
-                hset_contains(tab, tmask, (uint32_t)cand[t][k]))
-              wc[t][k] = 0.f;
-            partial += wc[t][k];
-          }
-          const float tot = wave_sum_uni(partial);
-          const uint64_t r = sm64_next(state[t]);  // drawn even on dead end
-          if (!(tot > 0.f)) { alive[t] = false; continue; }
-          const float target = (float)(u01_from(r) * (double)tot);
-          int chosen_cand = -1;
-          float base = 0.f;
-          int last_pos_cand = -1;
-          // fully unrolled with uniform guards: a runtime-indexed chunk
-          // loop sends cand/wc to scratch (rule-20 trap; measured 3x)
-#pragma unroll
-          for (int k = 0; k < WCHUNKS; ++k) {
-            if (k < nchunk && chosen_cand < 0) {
-              const float w = wc[t][k];
-              const float scan = wave_incl_scan(w);
-              const float chunk_tot = unif(__shfl(scan, WAVE - 1));
-              const bool hit = (w > 0.f) && (base + scan > target) &&
-                               (base + scan - w <= target);
-              const unsigned long long m = __ballot(hit);
-              if (m != 0ULL) {
-                chosen_cand = uni(__shfl(cand[t][k],
-                                         __ffsll((long long)m) - 1));
-              } else {
-                const unsigned long long mp = __ballot(w > 0.f);
-                if (mp != 0ULL)
-                  last_pos_cand = uni(__shfl(cand[t][k],
-                                             63 - __clzll((long long)mp)));
-                base += chunk_tot;
-              }
-            }
-          }
-          if (chosen_cand < 0) {
-            // rounding tail: target >= running total -> last unvisited
-            if (last_pos_cand < 0) { alive[t] = false; continue; }
-            chosen_cand = last_pos_cand;
-          }
-          cur[t] = chosen_cand;
-        } else {
-          // chunked fallback for very-high-degree rows (> 256 neighbors)
-          const int s = rs[t], dg = deg[t];
-          float partial = 0.f;
-          for (int j = lane; j < dg; j += WAVE) {
+          if (k < nchunk && j < deg) {
             const int cc = col_idx[s + j];
+            creg[k] = cc;
             float w = wgt[s + j];
             if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
+            wreg[k] = w;
             partial += w;
           }
-          const float tot = wave_sum_uni(partial);
-          const uint64_t r = sm64_next(state[t]);
-          if (!(tot > 0.f)) { alive[t] = false; continue; }
-          const float target = (float)(u01_from(r) * (double)tot);
-          int chosen = -1;
-          float base = 0.f;
-          for (int j0 = 0; j0 < dg; j0 += WAVE) {
+        }
+        const float tot = wave_sum_uni(partial);
+        const uint64_t r = sm64_next(state);   // drawn even on dead end
+        if (!(tot > 0.f)) break;
+        const float target = (float)(u01_from(r) * (double)tot);
+        int chosen_cand = -1;
+        float base = 0.f;
+        unsigned long long any_pos = 0ULL;
+        int last_pos_cand = -1;
+        for (int k = 0; k < nchunk; ++k) {
+          const float w = wreg[k];
+          const float scan = wave_incl_scan(w);
+          const float chunk_tot = unif(__shfl(scan, WAVE - 1));
+          const bool hit = (w > 0.f) && (base + scan > target) &&
+                           (base + scan - w <= target);
+          const unsigned long long m = __ballot(hit);
+          if (m != 0ULL) {
+            chosen_cand = uni(__shfl(creg[k], __ffsll((long long)m) - 1));
+            break;
+          }
+          const unsigned long long mp = __ballot(w > 0.f);
+          if (mp != 0ULL) {
+            any_pos = 1;
+            last_pos_cand = uni(__shfl(creg[k], 63 - __clzll((long long)mp)));
+          }
+          base += chunk_tot;
+        }
+        if (chosen_cand < 0) {
+          // rounding tail: target >= running total -> last unvisited
+          if (!any_pos) break;                 // cannot happen when tot > 0
+          chosen_cand = last_pos_cand;
+        }
+        cur = chosen_cand;
+      } else {
+        // chunked fallback for very-high-degree nodes (> 256 neighbors)
+        float partial = 0.f;
+        for (int j = lane; j < deg; j += WAVE) {
+          const int cc = col_idx[s + j];
+          float w = wgt[s + j];
+          if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
+          partial += w;
+        }
+        const float tot = wave_sum_uni(partial);
+        const uint64_t r = sm64_next(state);
+        if (!(tot > 0.f)) break;
+        const float target = (float)(u01_from(r) * (double)tot);
+        int chosen = -1;
+        float base = 0.f;
+        for (int j0 = 0; j0 < deg; j0 += WAVE) {
+          const int j = j0 + lane;
+          float w = 0.f;
+          if (j < deg) {
+            const int cc = col_idx[s + j];
+            w = wgt[s + j];
+            if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
+          }
+          const float scan = wave_incl_scan(w);
+          const float chunk_tot = unif(__shfl(scan, WAVE - 1));
+          const bool hit = (j < deg) && (w > 0.f) &&
+                           (base + scan > target) && (base + scan - w <= target);
+          const unsigned long long m = __ballot(hit);
+          if (m != 0ULL) { chosen = j0 + (__ffsll((long long)m) - 1); break; }
+          base += chunk_tot;
+        }
+        if (chosen < 0) {
+          for (int j0 = ((deg - 1) / WAVE) * WAVE; j0 >= 0 && chosen < 0;
+               j0 -= WAVE) {
             const int j = j0 + lane;
             float w = 0.f;
-            if (j < dg) {
+            if (j < deg) {
               const int cc = col_idx[s + j];
               w = wgt[s + j];
               if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
             }
-            const float scan = wave_incl_scan(w);
-            const float chunk_tot = unif(__shfl(scan, WAVE - 1));
-            const bool hit = (j < dg) && (w > 0.f) &&
-                             (base + scan > target) &&
-                             (base + scan - w <= target);
-            const unsigned long long m = __ballot(hit);
-            if (m != 0ULL) { chosen = j0 + (__ffsll((long long)m) - 1); break; }
-            base += chunk_tot;
+            const unsigned long long m = __ballot(w > 0.f);
+            if (m != 0ULL) chosen = j0 + (63 - __clzll((long long)m));
           }
-          if (chosen < 0) {
-            for (int j0 = ((dg - 1) / WAVE) * WAVE; j0 >= 0 && chosen < 0;
-                 j0 -= WAVE) {
-              const int j = j0 + lane;
-              float w = 0.f;
-              if (j < dg) {
-                const int cc = col_idx[s + j];
-                w = wgt[s + j];
-                if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
-              }
-              const unsigned long long m = __ballot(w > 0.f);
-              if (m != 0ULL) chosen = j0 + (63 - __clzll((long long)m));
-            }
-            if (chosen < 0) { alive[t] = false; continue; }
-          }
-          cur[t] = uni(col_idx[s + chosen]);
+          if (chosen < 0) break;               // cannot happen when tot > 0
         }
+        cur = uni(col_idx[s + chosen]);
       }
     }
 
-    // ---- outputs for both slots
-#pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      const long long wk = base0 + t;
-      if (wk >= n_walks) continue;
-      const long long outb = wk * (long long)len_path;
-      for (int k = lane; k < len_path; k += WAVE)
-        out_nodes[outb + k] = (k < plen[t]) ? vis0[t * len_path + k] : -1;
-      if (lane == 0) {
-        out_len[wk] = plen[t];
-        out_hash[wk] = (long long)hash[t];
-      }
+    long long outb = walk * (long long)len_path;
+    for (int k = lane; k < len_path; k += WAVE)
+      out_nodes[outb + k] = (k < plen) ? vis[k] : -1;
+    if (lane == 0) {
+      out_len[walk] = plen;
+      out_hash[walk] = (long long)hash;
     }
   }
 }
