@@ -229,3 +229,28 @@ def test_gemv_kernels_match_torch(G, h):
     g = torch.empty(h, device=DEV)
     ops.gemv_cols(W, c, g)
     assert torch.allclose(g, torch.mv(W.t(), c), atol=1e-3)
+
+
+def test_walks_bitwise_high_degree_paths():
+    """Exercises all three sampling paths (lean <=64, register multichunk
+    <=256, chunked fallback >256) bitwise vs the CPU oracle: a dense graph
+    with a >256-degree hub, uniform (exactly representable) weights."""
+    rng = np.random.default_rng(42)
+    G = 400
+    edges = set()
+    for d in range(G):              # hub 0: degree ~G-1 (> 256)
+        if d != 0:
+            edges.add((0, d))
+    for s in range(1, G):           # mid-degree rows (~100)
+        for d in rng.choice(G, size=100, replace=False):
+            if int(d) != s:
+                edges.add((s, int(d)))
+    edges = np.array(sorted(edges))
+    rp, ci, w = _csr(edges, np.ones(len(edges)), G)
+    src = torch.arange(G, dtype=torch.int32)
+    cn, cl, ch = cpu_ref.random_walks(rp, ci, w, src, 2, 10, seed=77)
+    gn, gl, gh = ops.random_walks(rp.to(DEV), ci.to(DEV), w.to(DEV),
+                                  src.to(DEV), 2, 10, seed=77)
+    assert torch.equal(gl.cpu(), cl)
+    assert torch.equal(gn.cpu(), cn)
+    assert torch.equal(gh.cpu(), ch)
