@@ -145,14 +145,21 @@ def main() -> int:
     # ---- timed throughput region: fresh state, W warmup + K timed epochs
     st = trainer.setup(ps, pre_sharded=True)
     n_tr_global = trainer.n_tr_global
-    for _ in range(args.warmup):
-        trainer.run_epoch(st)
+    pipelined = on_gpu and args.trainer_path == "fast"
+    if pipelined:
+        trainer.run_epochs_pipelined(st, args.warmup, early_stop=False)
+    else:
+        for _ in range(args.warmup):
+            trainer.run_epoch(st)
     ctx.barrier()
     if on_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        trainer.run_epoch(st)
+    if pipelined:
+        trainer.run_epochs_pipelined(st, args.steps, early_stop=False)
+    else:
+        for _ in range(args.steps):
+            trainer.run_epoch(st)
     if on_gpu:
         torch.cuda.synchronize()
     ctx.barrier()

@@ -281,10 +281,96 @@ class CbowTrainer:
         st.epoch_idx += 1
         return acc_tr, acc_val
 
+    def _ensure_graph(self, st) -> None:
+        """Capture the epoch body into a hipGraph when eligible (see the
+        world==1 gate rationale in run_epoch)."""
+        if (self.cfg.use_hipgraph and self.ctx.world == 1 and
+                st.graph is None and not st.graph_failed and
+                st.epoch_idx >= 1):
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._epoch_body_fast(st)
+                st.graph = g            # capture records without executing
+            except Exception as e:  # noqa: BLE001
+                st.graph_failed = True
+                self.log(f"    (hipGraph capture unavailable: {e!r}; "
+                         f"running eager)")
+
+    def _launch_epoch(self, st, slot: int, pinned, snaps, events) -> None:
+        """Queue one whole epoch asynchronously: body (graph replay or
+        eager), D2H copy of the accuracy counts into pinned memory, rolling
+        weight snapshot, completion event. Stream order guarantees the copy
+        reads THIS epoch's counts before the next epoch's body overwrites
+        them."""
+        st.t_adam += 1
+        st.lrt_buf.fill_(ops.tf1_lr_t(self.cfg.lr, self.B1, self.B2,
+                                      st.t_adam))
+        self._ensure_graph(st)
+        if st.graph is not None:
+            st.graph.replay()
+        else:
+            self._epoch_body_fast(st)
+        pinned[slot].copy_(st.counts_buf, non_blocking=True)
+        snaps[slot][0].copy_(st.W)          # post-epoch weights (async)
+        snaps[slot][1].copy_(st.who)
+        events[slot].record()
+        st.epoch_idx += 1
+
+    def run_epochs_pipelined(self, st, n_epochs: int, early_stop: bool,
+                             on_epoch=None):
+        """Speculative epoch pipeline (GPU fast-path full-batch only):
+        epoch e+1's launch is queued before epoch e's accuracy readback, so
+        the per-epoch D2H sync overlaps the next epoch's compute. Reference
+        semantics are exact: the accuracy trajectory is identical, an
+        early-stop truncates the history at the dip, the returned weights
+        are the pre-dip epoch's (rolling 3-deep snapshots implement the
+        keep-last-good rule, G2Vec.py:276-283); at most one speculative
+        epoch's compute is discarded. Returns (hist, stop_epoch, final_W,
+        final_who, last_acc_tr)."""
+        DEPTH = 3
+        pinned = [torch.empty(2, dtype=torch.float32, pin_memory=True)
+                  for _ in range(DEPTH)]
+        snaps = [(torch.empty_like(st.W), torch.empty_like(st.who))
+                 for _ in range(DEPTH)]
+        events = [torch.cuda.Event() for _ in range(DEPTH)]
+        hist = []
+        before_val = -1.0
+        stop_epoch = -1
+        launched = 0
+        acc_tr = 0.0
+        self._launch_epoch(st, 0, pinned, snaps, events)
+        launched = 1
+        e = 0
+        while True:
+            if launched < n_epochs and launched - e < DEPTH - 1:
+                self._launch_epoch(st, launched % DEPTH, pinned, snaps,
+                                   events)
+                launched += 1
+            events[e % DEPTH].synchronize()
+            cc = pinned[e % DEPTH]
+            acc_tr = float(cc[0]) / max(self.n_tr_global, 1)
+            acc_val = float(cc[1]) / max(self.n_vl_global, 1)
+            hist.append(acc_val)
+            if on_epoch is not None:
+                on_epoch(e, acc_tr, acc_val)
+            if early_stop and acc_val < before_val:
+                stop_epoch = e - 1      # dip at e: report/return epoch e-1
+                return (hist, stop_epoch, snaps[(e - 1) % DEPTH][0],
+                        snaps[(e - 1) % DEPTH][1], acc_tr)
+            before_val = acc_val
+            e += 1
+            if e >= n_epochs:
+                return hist, -1, st.W, st.who, acc_tr
+
     # ------------------------------------------------------------------ train
     def train(self, ps: PathSet, pre_sharded: bool = False) -> TrainResult:
         cfg = self.cfg
         st = self.setup(ps, pre_sharded)
+
+        if (cfg.trainer_path != "general" and cfg.batch_size == 0 and
+                self.device.type == "cuda"):
+            return self._train_pipelined(st)
 
         before_val, before_tr = -1.0, -1.0
         stop_epoch = -1
@@ -326,6 +412,49 @@ class CbowTrainer:
                            acc_val=acc_val, acc_tr=acc_tr,
                            epochs_run=epochs_run, acc_val_history=acc_hist,
                            epoch_times_s=epoch_times, wall_to_acc_s=wall_to_acc)
+
+    def _train_pipelined(self, st) -> TrainResult:
+        """train() driver over the speculative epoch pipeline (GPU
+        fast-path full batch). Output lines and results match the
+        synchronous loop exactly."""
+        cfg = self.cfg
+        tr_hist: List[float] = []
+        epoch_times: List[float] = []
+        wall_box = [None]
+        t0_all = time.perf_counter()
+        blk = [time.perf_counter()]
+        last = [t0_all]
+
+        self.log("     Start training the modified CBOW with early stopping")
+
+        def on_epoch(e, a_tr, a_val):
+            now = time.perf_counter()
+            epoch_times.append(now - last[0])
+            last[0] = now
+            tr_hist.append(a_tr)
+            if wall_box[0] is None and a_val >= self.ACC_TARGET:
+                wall_box[0] = now - t0_all
+            if e % 5 == 0:
+                self.log("    - Epoch: %03d\tACC[val]=%.4f\tACC[tr]=%.4f (%.3f sec)"
+                         % (e, a_val, a_tr, now - blk[0]))
+                blk[0] = now
+
+        hist, stop_epoch, W_final, _who_final, _ltr = \
+            self.run_epochs_pipelined(st, cfg.epochs, cfg.early_stop, on_epoch)
+        epochs_run = len(hist)
+        if stop_epoch >= 0:
+            acc_val, acc_tr = hist[stop_epoch], tr_hist[stop_epoch]
+            self.log("    - Epoch(stop): %03d\tACC[val]=%.4f\tACC[tr]=%.4f (%.3f sec)"
+                     % (stop_epoch, acc_val, acc_tr,
+                        time.perf_counter() - blk[0]))
+        else:
+            acc_val, acc_tr = hist[-1], tr_hist[-1]
+        self.log("    Optimization Finish")
+        return TrainResult(W_ih=W_final, stop_epoch=stop_epoch,
+                           acc_val=acc_val, acc_tr=acc_tr,
+                           epochs_run=epochs_run, acc_val_history=hist,
+                           epoch_times_s=epoch_times,
+                           wall_to_acc_s=wall_box[0])
 
     # ------------------------------------------------------------------ steps
     def _slice(self, ps: PathSet, lo: int, hi: int):
