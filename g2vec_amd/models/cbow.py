@@ -329,11 +329,14 @@ class CbowTrainer:
         epoch's compute is discarded. Returns (hist, stop_epoch, final_W,
         final_who, last_acc_tr)."""
         DEPTH = 3
-        pinned = [torch.empty(2, dtype=torch.float32, pin_memory=True)
-                  for _ in range(DEPTH)]
-        snaps = [(torch.empty_like(st.W), torch.empty_like(st.who))
-                 for _ in range(DEPTH)]
-        events = [torch.cuda.Event() for _ in range(DEPTH)]
+        if getattr(st, "pipe_bufs", None) is None:
+            st.pipe_bufs = (
+                [torch.empty(2, dtype=torch.float32, pin_memory=True)
+                 for _ in range(DEPTH)],
+                [(torch.empty_like(st.W), torch.empty_like(st.who))
+                 for _ in range(DEPTH)],
+                [torch.cuda.Event() for _ in range(DEPTH)])
+        pinned, snaps, events = st.pipe_bufs
         hist = []
         before_val = -1.0
         stop_epoch = -1
