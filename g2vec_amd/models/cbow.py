@@ -297,7 +297,8 @@ class CbowTrainer:
                 self.log(f"    (hipGraph capture unavailable: {e!r}; "
                          f"running eager)")
 
-    def _launch_epoch(self, st, slot: int, pinned, snaps, events) -> None:
+    def _launch_epoch(self, st, slot: int, pinned, snaps, events,
+                      keep: bool = True) -> None:
         """Queue one whole epoch asynchronously: body (graph replay or
         eager), D2H copy of the accuracy counts into pinned memory, rolling
         weight snapshot, completion event. Stream order guarantees the copy
@@ -312,8 +313,9 @@ class CbowTrainer:
         else:
             self._epoch_body_fast(st)
         pinned[slot].copy_(st.counts_buf, non_blocking=True)
-        snaps[slot][0].copy_(st.W)          # post-epoch weights (async)
-        snaps[slot][1].copy_(st.who)
+        if keep:                            # rolling keep-last-good snapshot;
+            snaps[slot][0].copy_(st.W)      # skipped when early_stop is off
+            snaps[slot][1].copy_(st.who)    # (2 GB/epoch at the 1M x 512 cfg)
         events[slot].record()
         st.epoch_idx += 1
 
@@ -333,22 +335,26 @@ class CbowTrainer:
             st.pipe_bufs = (
                 [torch.empty(2, dtype=torch.float32, pin_memory=True)
                  for _ in range(DEPTH)],
-                [(torch.empty_like(st.W), torch.empty_like(st.who))
-                 for _ in range(DEPTH)],
+                ([(torch.empty_like(st.W), torch.empty_like(st.who))
+                  for _ in range(DEPTH)] if early_stop else None),
                 [torch.cuda.Event() for _ in range(DEPTH)])
         pinned, snaps, events = st.pipe_bufs
+        if early_stop and snaps is None:    # first call was early_stop=False
+            snaps = [(torch.empty_like(st.W), torch.empty_like(st.who))
+                     for _ in range(DEPTH)]
+            st.pipe_bufs = (pinned, snaps, events)
         hist = []
         before_val = -1.0
         stop_epoch = -1
         launched = 0
         acc_tr = 0.0
-        self._launch_epoch(st, 0, pinned, snaps, events)
+        self._launch_epoch(st, 0, pinned, snaps, events, keep=early_stop)
         launched = 1
         e = 0
         while True:
             if launched < n_epochs and launched - e < DEPTH - 1:
                 self._launch_epoch(st, launched % DEPTH, pinned, snaps,
-                                   events)
+                                   events, keep=early_stop)
                 launched += 1
             events[e % DEPTH].synchronize()
             cc = pinned[e % DEPTH]
