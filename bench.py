@@ -84,6 +84,9 @@ def main() -> int:
     ap.add_argument("--n-modules", type=int, default=16)
     ap.add_argument("--acc-target-epochs", type=int, default=60)
     ap.add_argument("--no-hipgraph", action="store_true")
+    ap.add_argument("--no-pipeline", action="store_true",
+                    help="time the synchronous run_epoch loop instead of "
+                         "run_epochs_pipelined (A/B diagnostics)")
     args = ap.parse_args()
 
     ctx = init_dist("auto")
@@ -145,7 +148,8 @@ def main() -> int:
     # ---- timed throughput region: fresh state, W warmup + K timed epochs
     st = trainer.setup(ps, pre_sharded=True)
     n_tr_global = trainer.n_tr_global
-    pipelined = on_gpu and args.trainer_path == "fast"
+    pipelined = (on_gpu and args.trainer_path == "fast"
+                 and not args.no_pipeline)
     if pipelined:
         trainer.run_epochs_pipelined(st, args.warmup, early_stop=False)
     else:
