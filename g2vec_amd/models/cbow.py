@@ -39,6 +39,17 @@ from ..parallel.dist import DistContext, single
 from ..paths import PathSet, subset
 
 
+class _CpuEvent:
+    """No-op stand-in for torch.cuda.Event: lets the epoch pipeline's
+    ordering/collective logic run (and be gloo-tested) on CPU hosts, where
+    every launch is synchronous anyway."""
+    def record(self):
+        pass
+
+    def synchronize(self):
+        pass
+
+
 @dataclasses.dataclass
 class TrainResult:
     W_ih: torch.Tensor            # f32 [G, h] on the training device
@@ -331,13 +342,15 @@ class CbowTrainer:
         epoch's compute is discarded. Returns (hist, stop_epoch, final_W,
         final_who, last_acc_tr)."""
         DEPTH = 3
+        on_gpu = self.device.type == "cuda"
         if getattr(st, "pipe_bufs", None) is None:
             st.pipe_bufs = (
-                [torch.empty(2, dtype=torch.float32, pin_memory=True)
+                [torch.empty(2, dtype=torch.float32, pin_memory=on_gpu)
                  for _ in range(DEPTH)],
                 ([(torch.empty_like(st.W), torch.empty_like(st.who))
                   for _ in range(DEPTH)] if early_stop else None),
-                [torch.cuda.Event() for _ in range(DEPTH)])
+                [torch.cuda.Event() if on_gpu else _CpuEvent()
+                 for _ in range(DEPTH)])
         pinned, snaps, events = st.pipe_bufs
         if early_stop and snaps is None:    # first call was early_stop=False
             snaps = [(torch.empty_like(st.W), torch.empty_like(st.who))

@@ -133,3 +133,47 @@ def test_dp2_pipeline_end_to_end(tiny_files, tmp_path):
     assert n_paths_dp == res["n_paths"]
     assert n_gip_dp == res["n_genes_in_paths"]
     assert abs(acc_dp - res["acc_val"]) < 0.05
+
+
+def _pipelined_epochs_worker(rank, world, port, out):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        ps = _pathset()
+        cfg = G2VecConfig(hidden=64, epochs=12, early_stop=True, seed=4,
+                          device="cpu", dtype="fp32")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                         log=lambda *a, **k: None)
+        res_sync = tr.train(ps)             # CPU train() is the sync loop
+        st = tr.setup(ps)
+        hist, stop, W, _who, _ = tr.run_epochs_pipelined(
+            st, cfg.epochs, early_stop=True)
+        # pipelined DP epochs == sync DP epochs, on every rank
+        assert hist == pytest.approx(res_sync.acc_val_history, abs=1e-6)
+        assert stop == res_sync.stop_epoch
+        assert torch.allclose(W, res_sync.W_ih, atol=1e-6)
+        if rank == 0:
+            out.put((np.asarray(W), hist))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_pipelined_epochs_match_sync():
+    """The speculative epoch pipeline under world_size=2 (the path the
+    multi-GPU SCALE bench runs, minus RCCL): per-epoch collectives issued
+    from queued epochs must give the sync loop's exact trajectory and
+    keep-last-good weights."""
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    out = ctxm.Queue()
+    procs = [ctxm.Process(target=_pipelined_epochs_worker,
+                          args=(r, 2, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    W_dp, hist_dp = out.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert len(hist_dp) >= 2 and np.isfinite(W_dp).all()

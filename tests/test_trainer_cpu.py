@@ -148,3 +148,24 @@ def test_minibatch_runs():
                      log=lambda *a, **k: None)
     res = tr.train(ps)
     assert res.epochs_run == 3 and np.isfinite(res.acc_val)
+
+
+def test_pipelined_epochs_match_sync_loop():
+    """run_epochs_pipelined (the GPU epoch pipeline, exercised here via its
+    CPU event stub) must produce the sync loop's exact trajectory, stop
+    epoch and keep-last-good weights — both with and without early stop."""
+    ps = _random_pathset(G=30, P=120, seed=7)
+    for early_stop in (False, True):
+        cfg = G2VecConfig(hidden=64, epochs=40, early_stop=early_stop,
+                          seed=1, device="cpu")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                         log=lambda *a, **k: None)
+        res_sync = tr.train(ps)                       # CPU train() is sync
+        st = tr.setup(ps)
+        hist, stop, W, who, _ = tr.run_epochs_pipelined(
+            st, cfg.epochs, early_stop=early_stop)
+        assert hist == pytest.approx(res_sync.acc_val_history, abs=1e-6)
+        assert stop == res_sync.stop_epoch
+        assert torch.allclose(W, res_sync.W_ih, atol=1e-6)
+        if early_stop and stop >= 0:
+            assert hist[stop + 1] < hist[stop]
