@@ -127,20 +127,10 @@ class CbowTrainer:
 
     @staticmethod
     def _locality_sort(ps: PathSet) -> PathSet:
-        """Order the full-batch shard by (path length, first gene).
-
-        Length-major: the fwd/eval kernels run one 16-lane sub-wave per
-        path and a wave retires at the max of its 4 sub-waves' lengths —
-        with the walk lengths' heavy tail, unsorted order wastes ~3x
-        (measured at the 1M config, tools/exp_gather_locality.py).
-        First-gene minor: co-module paths land in adjacent sub-waves so
-        their s/dO gathers hit cache-resident slices."""
         if ps.n_paths == 0:
             return ps
-        offs = ps.offsets.long()
-        first = ps.genes[offs[:-1]].long()
-        length = offs[1:] - offs[:-1]
-        order = torch.argsort(length * (ps.n_genes + 1) + first, stable=True)
+        first = ps.genes[ps.offsets[:-1].long()].long()
+        order = torch.argsort(first, stable=True)
         return subset(ps, order)
 
     # ------------------------------------------------------------------ setup
@@ -186,17 +176,10 @@ class CbowTrainer:
         # persistent fast-path buffers (stable addresses across hipGraph replays)
         st.s_buf = None
         st.gradwho_buf = torch.empty_like(who)
-        st.dOinst_buf = None
         if not use_general:
             st.s_buf = torch.empty(self.G, dtype=torch.float32,
                                    device=self.device)
             ops.gemv_rows(W, who, st.s_buf)
-            if self.device.type == "cuda" and cfg.batch_size == 0:
-                # fused-scatter staging: per-instance dO in gene-sorted
-                # slot order (written by fwd, streamed by the segment sum)
-                st.dOinst_buf = torch.empty(tr.genes.numel(),
-                                            dtype=torch.float32,
-                                            device=self.device)
         st.lrt_buf = torch.zeros(1, dtype=torch.float32, device=self.device)
         st.counts_buf = torch.zeros(2, dtype=torch.float32, device=self.device)
         st.graph = None
@@ -221,16 +204,9 @@ class CbowTrainer:
         tr, vl = st.tr, st.vl
         st.counts_buf.zero_()
         lrt = st.lrt_buf if self.device.type == "cuda" else None
-        if st.dOinst_buf is not None:
-            _loss, _corr, dO = ops.cbow_fwd_scalar(
-                st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True,
-                plan=st.plan, dO_inst=st.dOinst_buf)
-            c = ops.scatter_dO(tr.genes, tr.offsets, st.dOinst_buf, self.G,
-                               plan=st.plan, from_inst=True)
-        else:
-            _loss, _corr, dO = ops.cbow_fwd_scalar(
-                st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True)
-            c = ops.scatter_dO(tr.genes, tr.offsets, dO, self.G, plan=st.plan)
+        _loss, _corr, dO = ops.cbow_fwd_scalar(
+            st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True)
+        c = ops.scatter_dO(tr.genes, tr.offsets, dO, self.G, plan=st.plan)
         self.ctx.allreduce_(c)                  # C1: whole dW_ih message
         ops.gemv_cols(st.W, c, st.gradwho_buf)  # dW_ho = W_ih^T c (pre-update W)
         ops.adam_rank1(st.W, st.mW, st.vW, c, st.who, st.t_adam, cfg.lr,

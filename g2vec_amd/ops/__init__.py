@@ -46,21 +46,10 @@ def random_walks(row_ptr, col_idx, weights, sources, num_repetition: int,
 
 
 # ------------------------------------------------------------------ CBOW fast (scalar) path
-def cbow_fwd_scalar(s, genes, offsets, labels, inv_b: float, want_grad: bool,
-                    plan=None, dO_inst=None):
-    """plan+dO_inst (GPU): fuse the backward scatter into the forward —
-    each path's dO is written straight to its gene-sorted instance slots
-    (fire-and-forget random stores), so the subsequent scatter_dO segment
-    reduce streams contiguously instead of random-gathering dO per
-    instance (4.4 ms -> ~0.1 ms at the 1M-gene config)."""
+def cbow_fwd_scalar(s, genes, offsets, labels, inv_b: float, want_grad: bool):
     if s.is_cuda:
-        if plan is not None:
-            loss, correct, dO = native().cbow_fwd_scalar(
-                s, genes, offsets, labels, float(inv_b), bool(want_grad),
-                inv_slot=plan.inv_slot, dO_inst=dO_inst)
-        else:
-            loss, correct, dO = native().cbow_fwd_scalar(
-                s, genes, offsets, labels, float(inv_b), bool(want_grad))
+        loss, correct, dO = native().cbow_fwd_scalar(
+            s, genes, offsets, labels, float(inv_b), bool(want_grad))
         return loss, correct, (dO if want_grad else None)
     return cpu_ref.cbow_fwd_scalar(s, genes, offsets, labels, inv_b, want_grad)
 
@@ -86,7 +75,6 @@ class ScatterPlan(NamedTuple):
     inst_path: torch.Tensor   # i32 [nnz]  path id of each instance, gene-sorted
     seg_start: torch.Tensor   # i32 [n_seg+1]
     seg_gene: torch.Tensor    # i32 [n_seg]
-    inv_slot: torch.Tensor    # i32 [nnz]  path-order position -> sorted slot
 
 
 def build_scatter_plan(genes: torch.Tensor, offsets: torch.Tensor,
@@ -99,24 +87,17 @@ def build_scatter_plan(genes: torch.Tensor, offsets: torch.Tensor,
     seg_gene, seg_counts = torch.unique_consecutive(sorted_genes, return_counts=True)
     seg_start = torch.zeros(len(seg_gene) + 1, dtype=torch.int64, device=genes.device)
     torch.cumsum(seg_counts, 0, out=seg_start[1:])
-    inv_slot = torch.empty_like(perm, dtype=torch.int32)
-    inv_slot[perm] = torch.arange(perm.numel(), dtype=torch.int32,
-                                  device=genes.device)
     return ScatterPlan(inst_path, seg_start.int().contiguous(),
-                       seg_gene.int().contiguous(), inv_slot.contiguous())
+                       seg_gene.int().contiguous())
 
 
 def scatter_dO(genes, offsets, dO, n_genes: int,
-               plan: Optional[ScatterPlan] = None, from_inst: bool = False):
-    """from_inst=True: dO is already in gene-sorted instance order (the
-    fused cbow_fwd_scalar wrote it) — bitwise-identical result, contiguous
-    reads."""
+               plan: Optional[ScatterPlan] = None):
     if dO.is_cuda:
         if plan is None:
             plan = build_scatter_plan(genes, offsets, n_genes)
-        return native().scatter_dO_det(
-            None if from_inst else plan.inst_path, plan.seg_start,
-            plan.seg_gene, dO, int(n_genes))
+        return native().scatter_dO_det(plan.inst_path, plan.seg_start,
+                                       plan.seg_gene, dO, int(n_genes))
     return cpu_ref.scatter_dO(genes, offsets, dO, n_genes)
 
 

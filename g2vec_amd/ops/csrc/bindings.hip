@@ -5,7 +5,6 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
-#include <optional>
 #include <cstring>
 #include <fstream>
 #include <sstream>
@@ -76,17 +75,7 @@ std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col
 // ------------------------------------------------------------ CBOW (fast)
 std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
                                            torch::Tensor offs, torch::Tensor labels,
-                                           double inv_b, bool want_grad,
-                                           std::optional<torch::Tensor> inv_slot,
-                                           std::optional<torch::Tensor> dO_inst) {
-  const bool fuse_scat = inv_slot.has_value();
-  if (fuse_scat) {
-    CHECK_DEV(*inv_slot); CHECK_CONT(*inv_slot); CHECK_I32(*inv_slot);
-    CHECK_DEV(*dO_inst); CHECK_CONT(*dO_inst); CHECK_F32(*dO_inst);
-    TORCH_CHECK(inv_slot->numel() == genes.numel() &&
-                dO_inst->numel() == genes.numel(),
-                "inv_slot/dO_inst must have one entry per path-gene instance");
-  }
+                                           double inv_b, bool want_grad) {
   CHECK_DEV(s); CHECK_CONT(s); CHECK_F32(s);
   CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
   CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
@@ -105,9 +94,7 @@ std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
                      offs.data_ptr<int>(), labels.data_ptr<float>(), P,
                      (float)inv_b, loss.data_ptr<float>(),
                      correct.data_ptr<float>(),
-                     want_grad ? dO.data_ptr<float>() : nullptr,
-                     fuse_scat ? inv_slot->data_ptr<int>() : nullptr,
-                     fuse_scat ? dO_inst->data_ptr<float>() : nullptr);
+                     want_grad ? dO.data_ptr<float>() : nullptr);
   LAUNCH_CHECK();
   return {loss, correct, dO};
 }
@@ -138,14 +125,10 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   LAUNCH_CHECK();
 }
 
-torch::Tensor scatter_dO_det(std::optional<torch::Tensor> inst_path,
-                             torch::Tensor seg_start,
+torch::Tensor scatter_dO_det(torch::Tensor inst_path, torch::Tensor seg_start,
                              torch::Tensor seg_gene, torch::Tensor dO,
                              int64_t n_genes) {
-  // inst_path given: dO is per-path, gathered via the plan.  inst_path
-  // absent: dO is ALREADY in gene-sorted instance order (written by the
-  // fused cbow_fwd_scalar) and the reduce is a pure contiguous stream.
-  if (inst_path) { CHECK_DEV(*inst_path); CHECK_CONT(*inst_path); CHECK_I32(*inst_path); }
+  CHECK_DEV(inst_path); CHECK_CONT(inst_path); CHECK_I32(inst_path);
   CHECK_DEV(seg_start); CHECK_CONT(seg_start); CHECK_I32(seg_start);
   CHECK_DEV(seg_gene); CHECK_CONT(seg_gene); CHECK_I32(seg_gene);
   CHECK_DEV(dO); CHECK_CONT(dO); CHECK_F32(dO);
@@ -154,8 +137,7 @@ torch::Tensor scatter_dO_det(std::optional<torch::Tensor> inst_path,
   const long long n_seg = seg_gene.numel();
   if (n_seg == 0) return c;
   hipLaunchKernelGGL(scatter_do_det_kernel, dim3(grid_for(n_seg, 4)), dim3(256),
-                     0, cur_stream(),
-                     inst_path ? inst_path->data_ptr<int>() : nullptr,
+                     0, cur_stream(), inst_path.data_ptr<int>(),
                      seg_start.data_ptr<int>(), seg_gene.data_ptr<int>(),
                      (int)n_seg, dO.data_ptr<float>(), c.data_ptr<float>());
   LAUNCH_CHECK();
@@ -459,13 +441,8 @@ parse_expression_tsv(const std::string& path) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("random_walks", &random_walks, "CSR biased random walks (gfx950)");
-  m.def("cbow_fwd_scalar", &cbow_fwd_scalar, "scalar CBOW forward + loss",
-        py::arg("s"), py::arg("genes"), py::arg("offs"), py::arg("labels"),
-        py::arg("inv_b"), py::arg("want_grad"),
-        py::arg("inv_slot") = py::none(), py::arg("dO_inst") = py::none());
-  m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO",
-        py::arg("inst_path"), py::arg("seg_start"), py::arg("seg_gene"),
-        py::arg("dO"), py::arg("n_genes"));
+  m.def("cbow_fwd_scalar", &cbow_fwd_scalar, "scalar CBOW forward + loss");
+  m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO");
   m.def("cbow_eval_counts_", &cbow_eval_counts_,
         "fused train/val correct-count eval (in-place counts[2])");
   m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");

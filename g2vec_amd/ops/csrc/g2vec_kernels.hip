@@ -361,9 +361,7 @@ extern "C" __global__ void __launch_bounds__(256)
 cbow_fwd_scalar_kernel(const float* __restrict__ s, const int* __restrict__ genes,
                        const int* __restrict__ offs, const float* __restrict__ labels,
                        long long P, float inv_b, float* __restrict__ loss,
-                       float* __restrict__ correct, float* __restrict__ dO,
-                       const int* __restrict__ inv_slot,
-                       float* __restrict__ dO_inst) {
+                       float* __restrict__ correct, float* __restrict__ dO) {
   const int sublane = threadIdx.x & (SUBW - 1);
   const int subs_per_block = blockDim.x / SUBW;
   const int sub = threadIdx.x / SUBW;
@@ -372,21 +370,13 @@ cbow_fwd_scalar_kernel(const float* __restrict__ s, const int* __restrict__ gene
     const int lo = offs[p], hi = offs[p + 1];
     float partial = 0.f;
     for (int i = lo + sublane; i < hi; i += SUBW) partial += s[genes[i]];
-    const float o = subwave_sum16(partial);   // every sublane holds the sum
-    const float y = labels[p];
-    const float d = (1.f / (1.f + expf(-o)) - y) * inv_b;
+    const float o = subwave_sum16(partial);
     if (sublane == 0) {
+      const float y = labels[p];
       loss[p] = fmaxf(o, 0.f) - o * y + log1pf(expf(-fabsf(o)));
       correct[p] = ((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f;
-      if (dO) dO[p] = d;
+      if (dO) dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
     }
-    // fused backward scatter: fan this path's dO out to its gene-sorted
-    // instance slots NOW (random fire-and-forget 4 B stores), so the
-    // c = X^T dO segment reduce reads a CONTIGUOUS stream instead of doing
-    // 92M latency-bound random dO gathers (the 4.4 ms top kernel of the
-    // 1M-gene epoch, profiles/1m_kernel_stats.csv)
-    if (dO_inst)
-      for (int i = lo + sublane; i < hi; i += SUBW) dO_inst[inv_slot[i]] = d;
   }
 }
 
@@ -467,23 +457,13 @@ scatter_do_det_kernel(const int* __restrict__ inst_path,
     const int lo = seg_start[sidx], hi = seg_start[sidx + 1];
     float p0 = 0.f, p1 = 0.f, p2 = 0.f, p3 = 0.f;
     int i = lo + lane;
-    if (inst_path) {              // gather dO by path id (plain scatter_dO)
-      for (; i + 3 * WAVE < hi; i += 4 * WAVE) {
-        p0 += dO[inst_path[i]];
-        p1 += dO[inst_path[i + WAVE]];
-        p2 += dO[inst_path[i + 2 * WAVE]];
-        p3 += dO[inst_path[i + 3 * WAVE]];
-      }
-      for (; i < hi; i += WAVE) p0 += dO[inst_path[i]];
-    } else {                      // dO already in instance order (fused fwd):
-      for (; i + 3 * WAVE < hi; i += 4 * WAVE) {  // pure streaming reads,
-        p0 += dO[i];                              // bitwise-identical sum
-        p1 += dO[i + WAVE];                       // order to the gather path
-        p2 += dO[i + 2 * WAVE];
-        p3 += dO[i + 3 * WAVE];
-      }
-      for (; i < hi; i += WAVE) p0 += dO[i];
+    for (; i + 3 * WAVE < hi; i += 4 * WAVE) {
+      p0 += dO[inst_path[i]];
+      p1 += dO[inst_path[i + WAVE]];
+      p2 += dO[inst_path[i + 2 * WAVE]];
+      p3 += dO[inst_path[i + 3 * WAVE]];
     }
+    for (; i < hi; i += WAVE) p0 += dO[inst_path[i]];
     const float v = wave_sum((p0 + p1) + (p2 + p3));
     if (lane == 0) c[seg_gene[sidx]] = v;
   }

@@ -121,26 +121,6 @@ def test_scatter_do_det_matches_oracle():
     assert torch.equal(c_gpu, c2)
 
 
-def test_fused_fwd_scatter_bitwise_equals_plain():
-    """The fused forward (dO written straight to gene-sorted instance
-    slots) + streaming segment sum must reproduce the plain
-    gather-by-path scatter BITWISE (same addends, same order)."""
-    genes, offs, labels = _pathset_tensors(seed=5)
-    s = torch.randn(300)
-    g, o, y = genes.to(DEV), offs.to(DEV), labels.to(DEV)
-    plan = ops.build_scatter_plan(g, o, 300)
-    l1, c1, dO = ops.cbow_fwd_scalar(s.to(DEV), g, o, y, 1 / 500, True)
-    c_plain = ops.scatter_dO(g, o, dO, 300, plan)
-    dO_inst = torch.empty(g.numel(), dtype=torch.float32, device=DEV)
-    l2, c2, dO2 = ops.cbow_fwd_scalar(s.to(DEV), g, o, y, 1 / 500, True,
-                                      plan=plan, dO_inst=dO_inst)
-    c_fused = ops.scatter_dO(g, o, dO_inst, 300, plan, from_inst=True)
-    assert torch.equal(l1, l2) and torch.equal(c1, c2)
-    assert torch.equal(dO, dO2)
-    assert torch.equal(dO_inst, dO2[plan.inst_path.long()])
-    assert torch.equal(c_plain, c_fused)
-
-
 def test_adam_kernels_match_oracle():
     torch.manual_seed(0)
     G, h = 128, 128
