@@ -71,24 +71,50 @@ def cbow_eval_counts_(s, genes, offsets, labels, p_split: int,
 class ScatterPlan(NamedTuple):
     """Precomputed gene-sorted instance layout for the deterministic
     c = X^T dO reduction (built once per path set; genes never change
-    across epochs)."""
+    across epochs).
+
+    When the dO table outgrows one XCD's 4 MB L2 (large path counts), the
+    instances are sorted by (path-slab, gene) instead — slab_seg_ptr marks
+    each ~3 MB slab's segment range and scatter_dO launches the reduce one
+    slab at a time, so every XCD gathers from an L2-resident dO slice.
+    Accumulation order (ascending slab) is fixed: still deterministic."""
     inst_path: torch.Tensor   # i32 [nnz]  path id of each instance, gene-sorted
     seg_start: torch.Tensor   # i32 [n_seg+1]
     seg_gene: torch.Tensor    # i32 [n_seg]
+    slab_seg_ptr: Optional[list] = None   # [n_slabs+1] segment-range bounds
+
+
+SLAB_BYTES = 3 << 20     # dO slice per launch; < the 4 MB per-XCD L2
 
 
 def build_scatter_plan(genes: torch.Tensor, offsets: torch.Tensor,
                        n_genes: int) -> ScatterPlan:
     counts = (offsets[1:] - offsets[:-1]).long()
+    P = len(counts)
     path_of = torch.repeat_interleave(
-        torch.arange(len(counts), device=genes.device), counts)
-    sorted_genes, perm = torch.sort(genes.long(), stable=True)
+        torch.arange(P, device=genes.device), counts)
+    slab_paths = SLAB_BYTES // 4
+    n_slabs = (P + slab_paths - 1) // slab_paths if genes.is_cuda else 1
+    if n_slabs <= 1:
+        sorted_key, perm = torch.sort(genes.long(), stable=True)
+    else:
+        key = (path_of // slab_paths) * n_genes + genes.long()
+        sorted_key, perm = torch.sort(key, stable=True)
     inst_path = path_of[perm].int().contiguous()
-    seg_gene, seg_counts = torch.unique_consecutive(sorted_genes, return_counts=True)
-    seg_start = torch.zeros(len(seg_gene) + 1, dtype=torch.int64, device=genes.device)
+    seg_key, seg_counts = torch.unique_consecutive(sorted_key,
+                                                   return_counts=True)
+    seg_start = torch.zeros(len(seg_key) + 1, dtype=torch.int64,
+                            device=genes.device)
     torch.cumsum(seg_counts, 0, out=seg_start[1:])
-    return ScatterPlan(inst_path, seg_start.int().contiguous(),
-                       seg_gene.int().contiguous())
+    if n_slabs <= 1:
+        return ScatterPlan(inst_path, seg_start.int().contiguous(),
+                           seg_key.int().contiguous())
+    seg_slab = seg_key // n_genes
+    seg_gene = (seg_key - seg_slab * n_genes).int().contiguous()
+    ptr = torch.searchsorted(
+        seg_slab, torch.arange(n_slabs + 1, device=genes.device)).cpu()
+    return ScatterPlan(inst_path, seg_start.int().contiguous(), seg_gene,
+                       [int(x) for x in ptr])
 
 
 def scatter_dO(genes, offsets, dO, n_genes: int,
@@ -96,8 +122,17 @@ def scatter_dO(genes, offsets, dO, n_genes: int,
     if dO.is_cuda:
         if plan is None:
             plan = build_scatter_plan(genes, offsets, n_genes)
-        return native().scatter_dO_det(plan.inst_path, plan.seg_start,
-                                       plan.seg_gene, dO, int(n_genes))
+        if plan.slab_seg_ptr is None:
+            return native().scatter_dO_det(plan.inst_path, plan.seg_start,
+                                           plan.seg_gene, dO, int(n_genes))
+        c = torch.zeros(n_genes, dtype=torch.float32, device=dO.device)
+        ptr = plan.slab_seg_ptr
+        for a, b in zip(ptr[:-1], ptr[1:]):     # ascending slabs: one
+            if b > a:                           # L2-resident slice at a time
+                native().scatter_dO_det_(plan.inst_path,
+                                         plan.seg_start[a:b + 1],
+                                         plan.seg_gene[a:b], dO, c)
+        return c
     return cpu_ref.scatter_dO(genes, offsets, dO, n_genes)
 
 

@@ -125,6 +125,27 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   LAUNCH_CHECK();
 }
 
+void scatter_dO_det_(torch::Tensor inst_path, torch::Tensor seg_start,
+                     torch::Tensor seg_gene, torch::Tensor dO,
+                     torch::Tensor c) {
+  // Accumulating out-arg variant: one slab's gene segments are reduced and
+  // ADDED into c (caller zeroes it once). The slab-blocked caller issues
+  // one launch per ~3 MB dO slice so each XCD's L2 holds the slice being
+  // gathered instead of thrashing across the whole dO table.
+  CHECK_DEV(inst_path); CHECK_CONT(inst_path); CHECK_I32(inst_path);
+  CHECK_DEV(seg_start); CHECK_CONT(seg_start); CHECK_I32(seg_start);
+  CHECK_DEV(seg_gene); CHECK_CONT(seg_gene); CHECK_I32(seg_gene);
+  CHECK_DEV(dO); CHECK_CONT(dO); CHECK_F32(dO);
+  CHECK_DEV(c); CHECK_CONT(c); CHECK_F32(c);
+  const long long n_seg = seg_gene.numel();
+  if (n_seg == 0) return;
+  hipLaunchKernelGGL(scatter_do_det_kernel, dim3(grid_for(n_seg, 4)), dim3(256),
+                     0, cur_stream(), inst_path.data_ptr<int>(),
+                     seg_start.data_ptr<int>(), seg_gene.data_ptr<int>(),
+                     (int)n_seg, dO.data_ptr<float>(), c.data_ptr<float>());
+  LAUNCH_CHECK();
+}
+
 torch::Tensor scatter_dO_det(torch::Tensor inst_path, torch::Tensor seg_start,
                              torch::Tensor seg_gene, torch::Tensor dO,
                              int64_t n_genes) {
@@ -443,6 +464,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("random_walks", &random_walks, "CSR biased random walks (gfx950)");
   m.def("cbow_fwd_scalar", &cbow_fwd_scalar, "scalar CBOW forward + loss");
   m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO");
+  m.def("scatter_dO_det_", &scatter_dO_det_,
+        "deterministic c += X^T dO (one slab, out-arg)");
   m.def("cbow_eval_counts_", &cbow_eval_counts_,
         "fused train/val correct-count eval (in-place counts[2])");
   m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");

@@ -465,7 +465,10 @@ scatter_do_det_kernel(const int* __restrict__ inst_path,
     }
     for (; i < hi; i += WAVE) p0 += dO[inst_path[i]];
     const float v = wave_sum((p0 + p1) + (p2 + p3));
-    if (lane == 0) c[seg_gene[sidx]] = v;
+    if (lane == 0) c[seg_gene[sidx]] += v;   // += so slab-blocked passes
+                                             // accumulate (c starts zeroed;
+                                             // launches are stream-ordered,
+                                             // so still deterministic)
   }
 }
 

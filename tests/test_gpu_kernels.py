@@ -121,6 +121,29 @@ def test_scatter_do_det_matches_oracle():
     assert torch.equal(c_gpu, c2)
 
 
+def test_scatter_slab_blocked_matches_unblocked():
+    """Slab-blocked scatter (one launch per L2-sized dO slice, ascending
+    accumulate into c) must match the single-pass reduce and be
+    deterministic across runs."""
+    genes, offs, labels = _pathset_tensors(seed=9)
+    dO = torch.randn(500)
+    g, o, d = genes.to(DEV), offs.to(DEV), dO.to(DEV)
+    c_ref = ops.scatter_dO(g, o, d, 300)        # single-slab (500 paths)
+    orig = ops.SLAB_BYTES
+    try:
+        ops.SLAB_BYTES = 4 * 64                 # force ~8 slabs of 64 paths
+        plan = ops.build_scatter_plan(g, o, 300)
+        assert plan.slab_seg_ptr is not None and len(plan.slab_seg_ptr) > 2
+        c1 = ops.scatter_dO(g, o, d, 300, plan)
+        c2 = ops.scatter_dO(g, o, d, 300, plan)
+    finally:
+        ops.SLAB_BYTES = orig
+    assert torch.equal(c1, c2)                  # deterministic
+    assert torch.allclose(c1, c_ref, atol=1e-5)
+    assert torch.allclose(c1.cpu(), cpu_ref.scatter_dO(genes, offs, dO, 300),
+                          atol=1e-5)
+
+
 def test_adam_kernels_match_oracle():
     torch.manual_seed(0)
     G, h = 128, 128
