@@ -193,6 +193,18 @@ class CbowTrainer:
             st.ev_labels = torch.cat([tr.labels, vl.labels]).contiguous()
         else:
             st.ev_genes = None
+        # full-batch steady state fuses the forward into the PREVIOUS
+        # epoch's eval (same s, same train paths — the gather runs once per
+        # weight version): dO_buf carries the pending dlogits; seed it here
+        # from the initial weights (the one standalone forward of a run)
+        st.dO_buf = None
+        if not use_general and cfg.batch_size == 0:
+            st.dO_buf = torch.zeros(tr.n_paths, dtype=torch.float32,
+                                    device=self.device)
+            if tr.n_paths > 0:
+                _l, _c, d0 = ops.cbow_fwd_scalar(
+                    st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True)
+                st.dO_buf.copy_(d0)
         return st
 
     def _epoch_body_fast(self, st) -> None:
@@ -204,21 +216,25 @@ class CbowTrainer:
         tr, vl = st.tr, st.vl
         st.counts_buf.zero_()
         lrt = st.lrt_buf if self.device.type == "cuda" else None
-        _loss, _corr, dO = ops.cbow_fwd_scalar(
-            st.s_buf, tr.genes, tr.offsets, tr.labels, st.inv_b, True)
-        c = ops.scatter_dO(tr.genes, tr.offsets, dO, self.G, plan=st.plan)
+        # dO for THIS epoch's step was emitted by the previous epoch's
+        # fused eval (or the setup() seeding) — no standalone forward
+        c = ops.scatter_dO(tr.genes, tr.offsets, st.dO_buf, self.G,
+                           plan=st.plan)
         self.ctx.allreduce_(c)                  # C1: whole dW_ih message
         ops.gemv_cols(st.W, c, st.gradwho_buf)  # dW_ho = W_ih^T c (pre-update W)
         ops.adam_rank1(st.W, st.mW, st.vW, c, st.who, st.t_adam, cfg.lr,
                        self.B1, self.B2, self.EPS, lrt_buf=lrt)
         ops.adam_dense(st.who, st.mO, st.vO, st.gradwho_buf, st.t_adam,
                        cfg.lr, self.B1, self.B2, self.EPS, lrt_buf=lrt)
-        # post-update accuracy (reference order, G2Vec.py:264-267):
-        # one fused eval kernel over the concatenated train+val paths
+        # post-update accuracy (reference order, G2Vec.py:264-267): one
+        # fused eval kernel over the concatenated train+val paths, which
+        # ALSO emits the next epoch's train dlogits (post-update s is the
+        # next epoch's pre-update s — bitwise the same forward)
         ops.gemv_rows(st.W, st.who, st.s_buf)
         if st.ev_genes is not None:
             ops.cbow_eval_counts_(st.s_buf, st.ev_genes, st.ev_offsets,
-                                  st.ev_labels, tr.n_paths, st.counts_buf)
+                                  st.ev_labels, tr.n_paths, st.counts_buf,
+                                  dO=st.dO_buf, inv_b=st.inv_b)
         self.ctx.allreduce_(st.counts_buf)      # C3: one fused metric reduce
 
     def run_epoch(self, st) -> tuple:

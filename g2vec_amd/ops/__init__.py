@@ -55,17 +55,25 @@ def cbow_fwd_scalar(s, genes, offsets, labels, inv_b: float, want_grad: bool):
 
 
 def cbow_eval_counts_(s, genes, offsets, labels, p_split: int,
-                      counts: torch.Tensor) -> None:
+                      counts: torch.Tensor, dO: Optional[torch.Tensor] = None,
+                      inv_b: float = 1.0) -> None:
     """Accumulate the concatenated train+val correct counts into counts[2]
-    (caller zeroes it). One fused kernel on GPU; oracle math on CPU."""
+    (caller zeroes it). One fused kernel on GPU; oracle math on CPU.
+
+    dO given: also emit the train split's (p < p_split) dlogit into dO with
+    scale inv_b — the fused next-epoch forward (this eval's s is the next
+    epoch's pre-update s), bitwise-identical to a separate cbow_fwd_scalar
+    over the train paths."""
     if s.is_cuda:
         native().cbow_eval_counts_(s, genes, offsets, labels, int(p_split),
-                                   counts)
+                                   counts, dO=dO, inv_b=float(inv_b))
         return
-    _l, corr, _d = cpu_ref.cbow_fwd_scalar(s, genes, offsets, labels, 1.0,
-                                           False)
+    _l, corr, d = cpu_ref.cbow_fwd_scalar(s, genes, offsets, labels, inv_b,
+                                          dO is not None)
     counts[0] += corr[:p_split].sum()
     counts[1] += corr[p_split:].sum()
+    if dO is not None:
+        dO.copy_(d[:p_split])
 
 
 class ScatterPlan(NamedTuple):

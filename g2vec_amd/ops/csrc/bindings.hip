@@ -5,6 +5,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <optional>
 #include <cstring>
 #include <fstream>
 #include <sstream>
@@ -101,7 +102,13 @@ std::vector<torch::Tensor> cbow_fwd_scalar(torch::Tensor s, torch::Tensor genes,
 
 void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
                        torch::Tensor labels, int64_t p_split,
-                       torch::Tensor counts) {
+                       torch::Tensor counts,
+                       std::optional<torch::Tensor> dO, double inv_b) {
+  if (dO) {
+    CHECK_DEV(*dO); CHECK_CONT(*dO); CHECK_F32(*dO);
+    TORCH_CHECK(dO->numel() >= p_split,
+                "dO must cover the train split (p < p_split)");
+  }
   CHECK_DEV(s); CHECK_CONT(s); CHECK_F32(s);
   CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
   CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
@@ -118,7 +125,8 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
                      dim3(256), 0, cur_stream(), s.data_ptr<float>(),
                      genes.data_ptr<int>(), offs.data_ptr<int>(),
                      labels.data_ptr<float>(), P, (long long)p_split,
-                     partials.data_ptr<float>());
+                     partials.data_ptr<float>(),
+                     dO ? dO->data_ptr<float>() : nullptr, (float)inv_b);
   hipLaunchKernelGGL(fold_partials_kernel, dim3(1), dim3(256), 0,
                      cur_stream(), partials.data_ptr<float>(), grid,
                      counts.data_ptr<float>());
@@ -467,6 +475,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_dO_det_", &scatter_dO_det_,
         "deterministic c += X^T dO (one slab, out-arg)");
   m.def("cbow_eval_counts_", &cbow_eval_counts_,
+        py::arg("s"), py::arg("genes"), py::arg("offs"), py::arg("labels"),
+        py::arg("p_split"), py::arg("counts"),
+        py::arg("dO") = py::none(), py::arg("inv_b") = 1.0,
         "fused train/val correct-count eval (in-place counts[2])");
   m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");
   m.def("adam_dense", &adam_dense, "TF1 Adam, dense grad");

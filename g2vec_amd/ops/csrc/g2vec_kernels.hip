@@ -389,7 +389,8 @@ extern "C" __global__ void __launch_bounds__(256)
 cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ genes,
                         const int* __restrict__ offs, const float* __restrict__ labels,
                         long long P, long long p_split,
-                        float* __restrict__ partials) {
+                        float* __restrict__ partials,
+                        float* __restrict__ dO, float inv_b) {
   const int sublane = threadIdx.x & (SUBW - 1);
   const int subs_per_block = blockDim.x / SUBW;
   const int sub = threadIdx.x / SUBW;
@@ -401,8 +402,15 @@ cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ gen
     for (int i = lo + sublane; i < hi; i += SUBW) partial += s[genes[i]];
     const float o = subwave_sum16(partial);
     if (sublane == 0) {
-      const float corr = (((o > 0.f ? 1.f : 0.f) == labels[p]) ? 1.f : 0.f);
+      const float y = labels[p];
+      const float corr = (((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f);
       if (p < p_split) c0 += corr; else c1 += corr;
+      // fused next-epoch forward: this eval's s IS the next epoch's
+      // pre-update s, so emit the train-split dO here and the standalone
+      // forward kernel disappears from steady-state epochs (the s-gather
+      // over the train paths runs ONCE per weight version, not twice)
+      if (dO && p < p_split)
+        dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
     }
   }
   // fold the per-thread accumulators: sublane0 lanes hold the values;
