@@ -199,6 +199,8 @@ def main() -> int:
                 "num_repetition": args.reps,
                 "trainer_path": args.trainer_path,
                 "hipgraph_active": bool(getattr(st, "graph", None) is not None),
+                "pipe_pinned": (bool(st.pipe_bufs[0][0].is_pinned())
+                                if getattr(st, "pipe_bufs", None) else None),
                 "val_acc": round(acc_val, 4),
                 "wall_to_val_acc_0.88_s": (round(wall_to_acc, 4)
                                            if wall_to_acc else None),
