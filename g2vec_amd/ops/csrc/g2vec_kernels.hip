@@ -399,7 +399,8 @@ cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ gen
        p += (long long)gridDim.x * subs_per_block) {
     const int lo = offs[p], hi = offs[p + 1];
     float partial = 0.f;
-    for (int i = lo + sublane; i < hi; i += SUBW) partial += s[genes[i]];
+    for (int i = lo + sublane; i < hi; i += SUBW)
+      partial += s[__builtin_nontemporal_load(&genes[i])];
     const float o = subwave_sum16(partial);
     if (sublane == 0) {
       const float y = labels[p];
@@ -504,7 +505,11 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
     const int j4 = (int)(i % h4);
     const float cg = c[g];
     const f32x4 wj = who4[j4];
-    f32x4 mm = m4[i], vv = v4[i], ww = W4[i];
+    // W/m/v are pure streams (no reuse inside an epoch): nontemporal
+    // ld/st keeps them from evicting the gather tables in the XCD L2s
+    f32x4 mm = __builtin_nontemporal_load(&m4[i]);
+    f32x4 vv = __builtin_nontemporal_load(&v4[i]);
+    f32x4 ww = __builtin_nontemporal_load(&W4[i]);
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
       const float grad = cg * wj[k];
@@ -512,7 +517,9 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
       vv[k] = b2 * vv[k] + (1.f - b2) * grad * grad;
       ww[k] -= lr_t * mm[k] / (sqrtf(vv[k]) + eps);
     }
-    m4[i] = mm; v4[i] = vv; W4[i] = ww;
+    __builtin_nontemporal_store(mm, &m4[i]);
+    __builtin_nontemporal_store(vv, &v4[i]);
+    __builtin_nontemporal_store(ww, &W4[i]);
   }
 }
 
