@@ -96,13 +96,17 @@ SLAB_BYTES = 3 << 20     # dO slice per launch; < the 4 MB per-XCD L2
 
 
 def build_scatter_plan(genes: torch.Tensor, offsets: torch.Tensor,
-                       n_genes: int) -> ScatterPlan:
+                       n_genes: int, _force_slabs: bool = False
+                       ) -> ScatterPlan:
+    # _force_slabs: CPU tests exercise the slab geometry (the CPU compute
+    # path itself never consumes a slabbed plan)
     counts = (offsets[1:] - offsets[:-1]).long()
     P = len(counts)
     path_of = torch.repeat_interleave(
         torch.arange(P, device=genes.device), counts)
     slab_paths = SLAB_BYTES // 4
-    n_slabs = (P + slab_paths - 1) // slab_paths if genes.is_cuda else 1
+    n_slabs = ((P + slab_paths - 1) // slab_paths
+               if (genes.is_cuda or _force_slabs) else 1)
     if n_slabs <= 1:
         sorted_key, perm = torch.sort(genes.long(), stable=True)
     else:

@@ -52,3 +52,44 @@ def test_subset():
     assert sub.offsets.tolist() == [0, 3, 5]
     assert sub.genes.tolist() == [3, 4, 5, 0, 1]
     assert sub.labels.tolist() == [0., 0.]
+
+
+def test_scatter_plan_slab_geometry():
+    """Slab-blocked plan (ops.SLAB_BYTES-sized path slabs): segments are
+    (slab, gene)-sorted, slab_seg_ptr partitions them, every instance is
+    covered exactly once, and each slab's instances stay in its path
+    range."""
+    import numpy as np
+
+    import g2vec_amd.ops as ops
+    rng = np.random.default_rng(3)
+    P, G = 64, 37
+    lens = rng.integers(1, 9, size=P)
+    genes = torch.tensor(
+        np.concatenate([rng.choice(G, size=n, replace=False) for n in lens]),
+        dtype=torch.int32)
+    offs = torch.tensor(np.concatenate([[0], np.cumsum(lens)]),
+                        dtype=torch.int32)
+    orig = ops.SLAB_BYTES
+    try:
+        ops.SLAB_BYTES = 4 * 16      # 16-path slabs -> 4 slabs
+        plan = ops.build_scatter_plan(genes, offs, G, _force_slabs=True)
+    finally:
+        ops.SLAB_BYTES = orig
+    ptr = plan.slab_seg_ptr
+    assert ptr is not None and ptr[0] == 0 and ptr[-1] == plan.seg_gene.numel()
+    assert int(plan.seg_start[-1]) == genes.numel()
+    seen = 0
+    for slab, (a, b) in enumerate(zip(ptr[:-1], ptr[1:])):
+        gs = plan.seg_gene[a:b].tolist()
+        assert gs == sorted(gs)                      # gene-sorted per slab
+        for sidx in range(a, b):
+            lo, hi = int(plan.seg_start[sidx]), int(plan.seg_start[sidx + 1])
+            assert hi > lo
+            for i in range(lo, hi):
+                p = int(plan.inst_path[i])
+                assert p // 16 == slab               # instance in its slab
+                assert int(plan.seg_gene[sidx]) in genes[
+                    int(offs[p]):int(offs[p + 1])].tolist()
+            seen += hi - lo
+    assert seen == genes.numel()
