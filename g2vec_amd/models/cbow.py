@@ -207,15 +207,24 @@ class CbowTrainer:
                 st.dO_buf.copy_(d0)
         return st
 
-    def _epoch_body_fast(self, st) -> None:
+    def _epoch_body_fast(self, st, counts_out=None, lrt_slot=None) -> None:
         """One full-batch fast-path epoch as a capturable body: optimizer
         step at W_t, then post-update s + accuracy counts. All inputs and
         outputs live in persistent buffers (s_buf, lrt_buf, counts_buf) so
-        the body can be recorded once into a hipGraph and replayed."""
+        the body can be recorded once into a hipGraph and replayed.
+
+        counts_out/lrt_slot override the default buffers — the k-epoch
+        block graph records k bodies, each bound to its own slot of a
+        [k,2] counts buffer and a [k] lr_t buffer."""
         cfg = self.cfg
         tr, vl = st.tr, st.vl
-        st.counts_buf.zero_()
-        lrt = st.lrt_buf if self.device.type == "cuda" else None
+        if counts_out is None:
+            counts_out = st.counts_buf
+        counts_out.zero_()
+        if lrt_slot is not None:
+            lrt = lrt_slot
+        else:
+            lrt = st.lrt_buf if self.device.type == "cuda" else None
         # dO for THIS epoch's step was emitted by the previous epoch's
         # fused eval (or the setup() seeding) — no standalone forward
         c = ops.scatter_dO(tr.genes, tr.offsets, st.dO_buf, self.G,
@@ -233,9 +242,9 @@ class CbowTrainer:
         ops.gemv_rows(st.W, st.who, st.s_buf)
         if st.ev_genes is not None:
             ops.cbow_eval_counts_(st.s_buf, st.ev_genes, st.ev_offsets,
-                                  st.ev_labels, tr.n_paths, st.counts_buf,
+                                  st.ev_labels, tr.n_paths, counts_out,
                                   dO=st.dO_buf, inv_b=st.inv_b)
-        self.ctx.allreduce_(st.counts_buf)      # C3: one fused metric reduce
+        self.ctx.allreduce_(counts_out)         # C3: one fused metric reduce
 
     def run_epoch(self, st) -> tuple:
         """One reference epoch: optimizer step(s) at W_t, then post-update
@@ -346,6 +355,96 @@ class CbowTrainer:
         events[slot].record()
         st.epoch_idx += 1
 
+    KBLOCK = 8   # epochs recorded per block graph (fixed-epoch runs)
+
+    def _run_epochs_kblocked(self, st, n_epochs: int, on_epoch):
+        """Fixed-epoch fast path (early stop OFF): KBLOCK epochs are
+        recorded into ONE hipGraph — each recorded body bound to its own
+        lr_t slot and [2]-counts slot — so host cost per epoch drops to
+        1/KBLOCK of a launch + one [K,2] pinned read per block. Exact
+        per-epoch semantics: every epoch still runs the full optimizer
+        pass and both accuracy evals, and every epoch's accuracy is read
+        on host in order. Returns the same tuple as run_epochs_pipelined.
+
+        Tiny-epoch configs are host-launch-bound even pipelined (~6 torch
+        calls per 35 us of GPU work at the bundled scale) — this is the
+        next rung of the same ladder."""
+        K = self.KBLOCK
+        DEPTH = 3
+        if st.epoch_idx == 0:    # one eager epoch: warm allocator/state
+            hist = [self.run_epoch(st)]
+        else:
+            hist = []
+        if getattr(st, "kbufs", None) is None:
+            st.kbufs = (
+                torch.empty(K, dtype=torch.float32, device=self.device),
+                torch.zeros(K, 2, dtype=torch.float32, device=self.device),
+                [torch.empty(K, 2, dtype=torch.float32, pin_memory=True)
+                 for _ in range(DEPTH)],
+                [torch.cuda.Event() for _ in range(DEPTH)])
+            st.kgraph = None
+        klrt, kcounts, pinned, events = st.kbufs
+        if st.kgraph is None:
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    for j in range(K):
+                        self._epoch_body_fast(st, counts_out=kcounts[j],
+                                              lrt_slot=klrt[j:j + 1])
+                st.kgraph = g        # capture records without executing
+            except Exception as e:  # noqa: BLE001
+                # capture mutates no epoch state (recording only); the
+                # tail loop below completes every remaining epoch eagerly
+                st.kgraph_failed = True
+                self.log(f"    (k-epoch block graph unavailable: {e!r}; "
+                         f"running per-epoch)")
+
+        n_blocks = ((n_epochs - len(hist)) // K
+                    if st.kgraph is not None else 0)
+        # whole lr_t schedule staged to device ONCE; each block slices it
+        # with a stream-ordered D2D copy (a host-side refill of klrt could
+        # race a replay still queued behind it)
+        sched = torch.tensor(
+            [ops.tf1_lr_t(self.cfg.lr, self.B1, self.B2, st.t_adam + i)
+             for i in range(1, n_blocks * K + 1)],
+            dtype=torch.float32, device=self.device)
+
+        def launch_block(b, slot):
+            klrt.copy_(sched[b * K:(b + 1) * K], non_blocking=True)
+            st.t_adam += K
+            st.epoch_idx += K
+            st.kgraph.replay()
+            pinned[slot].copy_(kcounts, non_blocking=True)
+            events[slot].record()
+        acc_tr = hist[-1][0] if hist else 0.0
+        hist = [h[1] for h in hist]
+        if on_epoch is not None and hist:
+            on_epoch(0, acc_tr, hist[0])
+        launched = 0
+        if n_blocks > 0:
+            launch_block(0, 0)
+            launched = 1
+        b = 0
+        while b < n_blocks:
+            if launched < n_blocks and launched - b < DEPTH - 1:
+                launch_block(launched, launched % DEPTH)
+                launched += 1
+            events[b % DEPTH].synchronize()
+            cc = pinned[b % DEPTH]
+            for j in range(K):
+                acc_tr = float(cc[j, 0]) / max(self.n_tr_global, 1)
+                acc_val = float(cc[j, 1]) / max(self.n_vl_global, 1)
+                if on_epoch is not None:
+                    on_epoch(len(hist), acc_tr, acc_val)
+                hist.append(acc_val)
+            b += 1
+        while len(hist) < n_epochs:            # tail epochs, one at a time
+            acc_tr, acc_val = self.run_epoch(st)
+            if on_epoch is not None:
+                on_epoch(len(hist), acc_tr, acc_val)
+            hist.append(acc_val)
+        return hist, -1, st.W, st.who, acc_tr
+
     def run_epochs_pipelined(self, st, n_epochs: int, early_stop: bool,
                              on_epoch=None):
         """Speculative epoch pipeline (GPU fast-path full-batch only):
@@ -359,6 +458,11 @@ class CbowTrainer:
         final_who, last_acc_tr)."""
         DEPTH = 3
         on_gpu = self.device.type == "cuda"
+        if (not early_stop and on_gpu and self.cfg.use_hipgraph and
+                self.ctx.world == 1 and not st.graph_failed and
+                not getattr(st, "kgraph_failed", False) and
+                st.ev_genes is not None and n_epochs >= 2 * self.KBLOCK):
+            return self._run_epochs_kblocked(st, n_epochs, on_epoch)
         if getattr(st, "pipe_bufs", None) is None:
             st.pipe_bufs = (
                 [torch.empty(2, dtype=torch.float32, pin_memory=on_gpu)

@@ -106,3 +106,39 @@ def test_hipgraph_matches_eager(tmp_path):
         weights[graphed] = res.W_ih
     assert hists[True] == hists[False]
     assert torch.equal(weights[True], weights[False])
+
+
+@pytest.mark.gpu
+def test_kblock_epochs_match_sync_loop(tmp_path):
+    """The k-epoch block graph (fixed-epoch fast path) must reproduce the
+    synchronous per-epoch loop's accuracy trajectory exactly."""
+    from g2vec_amd.models.cbow import CbowTrainer
+    from g2vec_amd.paths import PathSet
+    import numpy as np
+
+    rng = np.random.default_rng(11)
+    G, P = 400, 3000
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(2, 14))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda")
+    ps = PathSet(torch.tensor(genes, dtype=torch.int32, device=dev),
+                 torch.tensor(offs, dtype=torch.int32, device=dev),
+                 torch.tensor(labels, device=dev), G)
+    N_EP = 25                      # >= 2*KBLOCK -> k-block path
+    cfg = G2VecConfig(hidden=64, epochs=N_EP, early_stop=False, seed=3,
+                      device="cuda")
+    tr1 = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None)
+    st1 = tr1.setup(ps)
+    hist_sync = [tr1.run_epoch(st1)[1] for _ in range(N_EP)]
+    tr2 = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None)
+    st2 = tr2.setup(ps)
+    hist_k, stop, W, _who, _ = tr2.run_epochs_pipelined(
+        st2, N_EP, early_stop=False)
+    assert getattr(st2, "kgraph", None) is not None, "k-block graph not used"
+    assert stop == -1 and len(hist_k) == N_EP
+    assert hist_k == pytest.approx(hist_sync, abs=1e-6)
+    assert torch.allclose(W, st1.W, atol=1e-6)
