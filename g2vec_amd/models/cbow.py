@@ -367,10 +367,7 @@ class CbowTrainer:
         if getattr(st, "kbufs", None) is None:
             st.kbufs = (
                 torch.empty(K, dtype=torch.float32, device=self.device),
-                torch.zeros(K, 2, dtype=torch.float32, device=self.device),
-                [torch.empty(K, 2, dtype=torch.float32, pin_memory=True)
-                 for _ in range(3)],
-                [torch.cuda.Event() for _ in range(3)])
+                torch.zeros(K, 2, dtype=torch.float32, device=self.device))
             st.kgraph = None
         if st.kgraph is None and not getattr(st, "kgraph_failed", False):
             klrt, kcounts = st.kbufs[0], st.kbufs[1]
@@ -398,65 +395,58 @@ class CbowTrainer:
     def _run_epochs_kblocked(self, st, n_epochs: int, on_epoch):
         """Fixed-epoch fast path (early stop OFF): KBLOCK epochs are
         recorded into ONE hipGraph — each recorded body bound to its own
-        lr_t slot and [2]-counts slot — so host cost per epoch drops to
-        1/KBLOCK of a launch + one [K,2] pinned read per block. Exact
-        per-epoch semantics: every epoch still runs the full optimizer
-        pass and both accuracy evals, and every epoch's accuracy is read
-        on host in order. Returns the same tuple as run_epochs_pipelined.
-
-        Tiny-epoch configs are host-launch-bound even pipelined (~6 torch
-        calls per 35 us of GPU work at the bundled scale) — this is the
-        next rung of the same ladder."""
+        lr_t slot and [2]-counts slot. Per-epoch counts are appended to a
+        DEVICE-side history with D2D copies and the whole history is read
+        back in a single D2H at the end: zero mid-run host<->device
+        crossings. (A rare per-process driver slow-mode was traced to
+        degraded D2H copies — ~1 ms each regardless of size; with one
+        final read the worst case costs ~1 ms per RUN, not per block.)
+        Exact per-epoch semantics: every epoch runs the full optimizer
+        pass and both accuracy evals; fixed-epoch runs consume the
+        accuracies only after the loop, so deferring the readback changes
+        nothing observable. Returns the run_epochs_pipelined tuple."""
         K = self.KBLOCK
-        DEPTH = 3
         if st.epoch_idx == 0:    # one eager epoch: warm allocator/state
-            hist = [self.run_epoch(st)]
+            warm = [self.run_epoch(st)]
         else:
-            hist = []
+            warm = []
         self._ensure_kgraph(st)
-        klrt, kcounts, pinned, events = st.kbufs
+        klrt, kcounts = st.kbufs[0], st.kbufs[1]
 
-        n_blocks = ((n_epochs - len(hist)) // K
-                    if st.kgraph is not None else 0)
+        n_rest = n_epochs - len(warm)
+        n_blocks = (n_rest // K) if st.kgraph is not None else 0
+        n_tail = n_rest - n_blocks * K
         # whole lr_t schedule staged to device ONCE; each block slices it
         # with a stream-ordered D2D copy (a host-side refill of klrt could
         # race a replay still queued behind it)
         sched = torch.tensor(
             [ops.tf1_lr_t(self.cfg.lr, self.B1, self.B2, st.t_adam + i)
-             for i in range(1, n_blocks * K + 1)],
+             for i in range(1, n_rest + 1)],
             dtype=torch.float32, device=self.device)
-
-        def launch_block(b, slot):
+        hist_dev = torch.empty(max(n_rest, 1), 2, dtype=torch.float32,
+                               device=self.device)
+        for b in range(n_blocks):
             klrt.copy_(sched[b * K:(b + 1) * K], non_blocking=True)
             st.t_adam += K
             st.epoch_idx += K
             st.kgraph.replay()
-            pinned[slot].copy_(kcounts, non_blocking=True)
-            events[slot].record()
-        acc_tr = hist[-1][0] if hist else 0.0
-        hist = [h[1] for h in hist]
+            hist_dev[b * K:(b + 1) * K].copy_(kcounts, non_blocking=True)
+        for j in range(n_tail):                # tail: eager bodies, same
+            e = n_blocks * K + j               # zero-readback scheme
+            st.t_adam += 1
+            st.epoch_idx += 1
+            self._epoch_body_fast(st, counts_out=st.counts_buf,
+                                  lrt_slot=sched[e:e + 1])
+            hist_dev[e].copy_(st.counts_buf, non_blocking=True)
+
+        cc = hist_dev.cpu()                    # the ONE host readback
+        acc_tr = warm[-1][0] if warm else 0.0
+        hist = [h[1] for h in warm]
         if on_epoch is not None and hist:
             on_epoch(0, acc_tr, hist[0])
-        launched = 0
-        if n_blocks > 0:
-            launch_block(0, 0)
-            launched = 1
-        b = 0
-        while b < n_blocks:
-            if launched < n_blocks and launched - b < DEPTH - 1:
-                launch_block(launched, launched % DEPTH)
-                launched += 1
-            events[b % DEPTH].synchronize()
-            cc = pinned[b % DEPTH]
-            for j in range(K):
-                acc_tr = float(cc[j, 0]) / max(self.n_tr_global, 1)
-                acc_val = float(cc[j, 1]) / max(self.n_vl_global, 1)
-                if on_epoch is not None:
-                    on_epoch(len(hist), acc_tr, acc_val)
-                hist.append(acc_val)
-            b += 1
-        while len(hist) < n_epochs:            # tail epochs, one at a time
-            acc_tr, acc_val = self.run_epoch(st)
+        for e in range(n_rest):
+            acc_tr = float(cc[e, 0]) / max(self.n_tr_global, 1)
+            acc_val = float(cc[e, 1]) / max(self.n_vl_global, 1)
             if on_epoch is not None:
                 on_epoch(len(hist), acc_tr, acc_val)
             hist.append(acc_val)
