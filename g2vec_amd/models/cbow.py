@@ -262,23 +262,8 @@ class CbowTrainer:
             if on_gpu:
                 st.lrt_buf.fill_(ops.tf1_lr_t(cfg.lr, self.B1, self.B2,
                                               st.t_adam))
-            # capture only at world==1 for now: a replay-time RCCL issue
-            # inside a graph could hang rather than raise, and the eager
-            # fallback cannot catch that (collectives are latency-bound at
-            # this message size anyway). TODO round 2: validate RCCL+graphs
-            # on a real multi-GPU node and lift the gate.
-            if (on_gpu and cfg.use_hipgraph and self.ctx.world == 1 and
-                    st.graph is None and not st.graph_failed and
-                    st.epoch_idx >= 1):
-                try:
-                    g = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(g):
-                        self._epoch_body_fast(st)
-                    st.graph = g        # capture records without executing
-                except Exception as e:  # noqa: BLE001
-                    st.graph_failed = True
-                    self.log(f"    (hipGraph capture unavailable: {e!r}; "
-                             f"running eager)")
+            if on_gpu:
+                self._ensure_graph(st)
             if st.graph is not None:
                 st.graph.replay()
             else:
@@ -318,8 +303,12 @@ class CbowTrainer:
         return acc_tr, acc_val
 
     def _ensure_graph(self, st) -> None:
-        """Capture the epoch body into a hipGraph when eligible (see the
-        world==1 gate rationale in run_epoch)."""
+        """Capture the epoch body into a hipGraph when eligible. Capture
+        only at world==1 for now: a replay-time RCCL issue inside a graph
+        could hang rather than raise, and the eager fallback cannot catch
+        that (collectives are latency-bound at this message size anyway).
+        TODO round 2: validate RCCL+graphs on a real multi-GPU node and
+        lift the gate."""
         if (self.cfg.use_hipgraph and self.ctx.world == 1 and
                 st.graph is None and not st.graph_failed and
                 st.epoch_idx >= 1):
