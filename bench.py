@@ -200,7 +200,9 @@ def main() -> int:
                 "hidden": args.hidden,
                 "num_repetition": args.reps,
                 "trainer_path": args.trainer_path,
-                "hipgraph_active": bool(getattr(st, "graph", None) is not None),
+                "hipgraph_active": bool(
+                    getattr(st, "graph", None) is not None
+                    or getattr(st, "kgraph", None) is not None),
                 "pipe_pinned": (bool(st.pipe_bufs[0][0].is_pinned())
                                 if getattr(st, "pipe_bufs", None) else None),
                 "val_acc": round(acc_val, 4),
