@@ -1,7 +1,6 @@
 """Multi-process DP coverage on CPU (gloo, world_size=2): the C1-C5
 collective path must produce the same training result as single-process
 (SURVEY §4.4)."""
-import os
 import socket
 
 import numpy as np

@@ -7,7 +7,6 @@ import pytest
 import torch
 
 from g2vec_amd.config import G2VecConfig
-from g2vec_amd.io import load_expression
 from g2vec_amd.pipeline import run
 
 

@@ -1,6 +1,5 @@
 """Parser/writer round-trips against the documented formats."""
 import numpy as np
-import pytest
 
 from g2vec_amd.io import (load_clinical, load_expression, load_network,
                           write_biomarkers, write_lgroups, write_vectors)

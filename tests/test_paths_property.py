@@ -1,5 +1,4 @@
 """Property-based checks of path-set integration (hypothesis)."""
-import numpy as np
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st

@@ -7,7 +7,6 @@ import numpy as np
 import pytest
 import torch
 
-from g2vec_amd import ops
 from g2vec_amd.config import G2VecConfig
 from g2vec_amd.models.cbow import CbowTrainer, _trunc_normal
 from g2vec_amd.ops import cpu_ref
