@@ -77,7 +77,8 @@ def test_jsonl_metrics(tiny_files, tmp_path):
     log = str(tmp_path / "m.jsonl")
     run(_cfg(tiny_files, tmp_path, log_jsonl=log))
     import json
-    events = [json.loads(l) for l in open(log)]
+    with open(log) as fh:
+        events = [json.loads(l) for l in fh]
     names = {e["event"] for e in events}
     assert {"counts", "paths", "train", "phase"} <= names
 
