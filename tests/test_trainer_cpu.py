@@ -52,7 +52,7 @@ def test_rank1_gradients_match_autograd():
     inv_b = 1.0 / ps.n_paths
     lossv, correct, dO = cpu_ref.cbow_fwd_scalar(
         torch.mv(W.detach(), who.detach()), ps.genes, ps.offsets, y, inv_b, True)
-    assert abs(float(lossv.mean().detach()) - float(loss)) < 1e-5
+    assert abs(float(lossv.mean()) - float(loss.detach())) < 1e-5
     c = cpu_ref.scatter_dO(ps.genes, ps.offsets, dO, G)
     dW_fast = torch.outer(c, who.detach())
     dwho_fast = torch.mv(W.detach().t(), c)
