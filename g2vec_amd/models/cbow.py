@@ -351,7 +351,12 @@ class CbowTrainer:
         (recording executes nothing and mutates no epoch state). Callable
         ahead of time — bench.py warms it during the untimed warmup so the
         one-time capture never lands in a timed region. Requires at least
-        one prior eager epoch (st.epoch_idx >= 1)."""
+        one prior eager epoch (st.epoch_idx >= 1). Capture is gated to
+        world==1 (same RCCL-inside-graph rationale as _ensure_graph);
+        when gated off the caller's eager tail loop runs every epoch —
+        still with the deferred single readback."""
+        if self.ctx.world != 1 or not self.cfg.use_hipgraph:
+            return
         K = self.KBLOCK
         if getattr(st, "kbufs", None) is None:
             st.kbufs = (
@@ -375,11 +380,12 @@ class CbowTrainer:
                          f"running per-epoch)")
 
     def kblock_eligible(self, st, n_epochs: int, early_stop: bool) -> bool:
+        """Fixed-epoch runs go through the deferred-readback runner: block
+        graphs when world==1 allows capture, eager epoch bodies otherwise
+        (world>1 / --no-hipgraph) — either way zero mid-run D2H."""
         return (not early_stop and self.device.type == "cuda" and
-                self.cfg.use_hipgraph and self.ctx.world == 1 and
-                not st.graph_failed and
                 not getattr(st, "kgraph_failed", False) and
-                st.ev_genes is not None and n_epochs >= 2 * self.KBLOCK)
+                st.ev_genes is not None and n_epochs >= 2)
 
     def _run_epochs_kblocked(self, st, n_epochs: int, on_epoch):
         """Fixed-epoch fast path (early stop OFF): KBLOCK epochs are
@@ -400,11 +406,12 @@ class CbowTrainer:
         else:
             warm = []
         self._ensure_kgraph(st)
-        klrt, kcounts = st.kbufs[0], st.kbufs[1]
-
+        kgraph = getattr(st, "kgraph", None)
         n_rest = n_epochs - len(warm)
-        n_blocks = (n_rest // K) if st.kgraph is not None else 0
+        n_blocks = (n_rest // K) if kgraph is not None else 0
         n_tail = n_rest - n_blocks * K
+        if kgraph is not None:
+            klrt, kcounts = st.kbufs[0], st.kbufs[1]
         # whole lr_t schedule staged to device ONCE; each block slices it
         # with a stream-ordered D2D copy (a host-side refill of klrt could
         # race a replay still queued behind it)
@@ -418,7 +425,7 @@ class CbowTrainer:
             klrt.copy_(sched[b * K:(b + 1) * K], non_blocking=True)
             st.t_adam += K
             st.epoch_idx += K
-            st.kgraph.replay()
+            kgraph.replay()
             hist_dev[b * K:(b + 1) * K].copy_(kcounts, non_blocking=True)
         for j in range(n_tail):                # tail: eager bodies, same
             e = n_blocks * K + j               # zero-readback scheme
@@ -427,6 +434,8 @@ class CbowTrainer:
             self._epoch_body_fast(st, counts_out=st.counts_buf,
                                   lrt_slot=sched[e:e + 1])
             hist_dev[e].copy_(st.counts_buf, non_blocking=True)
+            if (j + 1) % 64 == 0:       # bound launch-queue depth on long
+                torch.cuda.synchronize()   # eager (world>1) runs
 
         cc = hist_dev.cpu()                    # the ONE host readback
         acc_tr = warm[-1][0] if warm else 0.0
