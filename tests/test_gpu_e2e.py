@@ -128,7 +128,8 @@ def test_kblock_epochs_match_sync_loop(tmp_path):
     ps = PathSet(torch.tensor(genes, dtype=torch.int32, device=dev),
                  torch.tensor(offs, dtype=torch.int32, device=dev),
                  torch.tensor(labels, device=dev), G)
-    N_EP = 25                      # >= 2*KBLOCK -> k-block path
+    N_EP = 29    # 1 warm + 3 blocks of 8 + 4 eager tail epochs:
+                 # covers the replay loop AND the deferred tail
     cfg = G2VecConfig(hidden=64, epochs=N_EP, early_stop=False, seed=3,
                       device="cuda")
     tr1 = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None)
