@@ -181,9 +181,10 @@ void adam_rank1(torch::Tensor W, torch::Tensor m, torch::Tensor v,
   CHECK_DEV(lrt); CHECK_F32(lrt);
   const long long G = W.size(0);
   const int h = (int)W.size(1);
-  TORCH_CHECK(h % 4 == 0, "hidden must be a multiple of 4");
-  const long long n4 = G * h / 4;
-  hipLaunchKernelGGL(adam_rank1_kernel, dim3(grid_for(n4, 256)), dim3(256), 0,
+  TORCH_CHECK(h % 4 == 0 && 256 % (h / 4) == 0 && h <= 1024,
+              "hidden must be a multiple of 4 with h/4 dividing 256");
+  const int rpb = 256 / (h / 4);   // rows per 256-thread block
+  hipLaunchKernelGGL(adam_rank1_kernel, dim3(grid_for(G, rpb)), dim3(256), 0,
                      cur_stream(), W.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), c.data_ptr<float>(),
                      who.data_ptr<float>(), G, h, lrt.data_ptr<float>(),

@@ -495,16 +495,20 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
                   float eps) {
   const float lr_t = lr_t_ptr[0];   // device-read so hipGraph replays see
                                     // the per-step bias-corrected value
-  const long long n4 = (long long)G * h / 4;
   f32x4* W4 = (f32x4*)W; f32x4* m4 = (f32x4*)m; f32x4* v4 = (f32x4*)v;
   const f32x4* who4 = (const f32x4*)who;
   const int h4 = h / 4;
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
-       i += (long long)gridDim.x * blockDim.x) {
-    const long long g = i / h4;
-    const int j4 = (int)(i % h4);
+  // row-major indexing: the flat-index variant paid a 64-bit div+mod per
+  // f32x4; here each thread owns a fixed column group and strides over
+  // rows (one div/mod per THREAD at setup, multiply-add per iteration)
+  const int rpb = blockDim.x / h4;             // rows per block (h <= 1024)
+  const int lrow = (int)threadIdx.x / h4;      // this thread's row-in-block
+  const int j4 = (int)threadIdx.x - lrow * h4; // and column group
+  const f32x4 wj = who4[j4];
+  for (long long g = (long long)blockIdx.x * rpb + lrow; g < G;
+       g += (long long)gridDim.x * rpb) {
+    const long long i = g * h4 + j4;
     const float cg = c[g];
-    const f32x4 wj = who4[j4];
     // W/m/v are pure streams (no reuse inside an epoch): nontemporal
     // ld/st keeps them from evicting the gather tables in the XCD L2s
     f32x4 mm = __builtin_nontemporal_load(&m4[i]);
