@@ -131,3 +131,18 @@ def test_save_model_checkpoint(tiny_files, tmp_path):
     assert blob["W_ih"].shape == (res["n_genes"], 128)
     assert blob["acc_val"] == res["acc_val"]
     assert len(blob["gene_index"]) == res["n_genes"]
+
+
+def test_config_validation_rejects_bad_values():
+    import pytest as _pt
+
+    from g2vec_amd.config import G2VecConfig
+
+    ok = dict(expression_file="e", clinical_file="c", network_file="n",
+              result_name="r")
+    G2VecConfig(**ok).validate()
+    for bad in (dict(hidden=100), dict(dtype="int8"), dict(len_path=0),
+                dict(len_path=1000), dict(pcc_mode="dense"),
+                dict(trainer_path="medium"), dict(kmeans_backend="cuml")):
+        with _pt.raises(ValueError):
+            G2VecConfig(**{**ok, **bad}).validate()
