@@ -51,6 +51,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph capture of the training epoch")
     p.add_argument("--save-paths", type=str, default="")
+    p.add_argument("--load-model", type=str, default="",
+                   help="skip training: load W_ih from a --save-model .pt "
+                        "(gene count and hidden size must match)")
     p.add_argument("--save-model", type=str, default="",
                    help="save trained weights + metadata as a .pt checkpoint")
     p.add_argument("--load-paths", type=str, default="")
@@ -71,6 +74,7 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, save_paths=a.save_paths,
         load_paths=a.load_paths, save_model=a.save_model,
+        load_model=a.load_model,
         log_jsonl=a.log_jsonl,
         use_hipgraph=not a.no_hipgraph)
 

@@ -38,6 +38,7 @@ class G2VecConfig:
                                     # "general": full gather/scatter kernel chain (K1-K8)
     save_paths: str = ""            # cache generated path set (de-facto checkpoint)
     save_model: str = ""            # save trained W_ih/W_ho + metadata (.pt)
+    load_model: str = ""            # resume: skip step 4, load W_ih from .pt
     load_paths: str = ""
     log_jsonl: str = ""             # structured metrics sink
     deterministic_grads: bool = False  # bitwise-reproducible dW_ih reduction (no atomics)

@@ -116,9 +116,27 @@ def run(cfg: G2VecConfig, ctx: Optional[DistContext] = None) -> Dict:
     jsonl.emit("paths", n_paths=ps.n_paths, n_genes_in_paths=n_in_paths, **stats)
 
     log(">>> 4. Compute distributed representations using modified CBOW")
-    trainer = CbowTrainer(cfg, n_genes, device, ctx, log=log)
-    with timers.phase("train"):
-        res = trainer.train(ps)
+    if cfg.load_model:
+        # resume from a --save-model checkpoint: skip training entirely
+        # (steps 5-7 only need W_ih); shape metadata must match this run
+        blob = torch.load(cfg.load_model, map_location="cpu")
+        if int(blob["n_genes"]) != n_genes or int(blob["hidden"]) != cfg.hidden:
+            raise ValueError(
+                f"--load-model checkpoint was trained on n_genes="
+                f"{blob['n_genes']}/hidden={blob['hidden']}, this run has "
+                f"n_genes={n_genes}/hidden={cfg.hidden}")
+        from .models.cbow import TrainResult
+        res = TrainResult(W_ih=blob["W_ih"].float(),
+                          stop_epoch=int(blob.get("stop_epoch", -1)),
+                          acc_val=float(blob.get("acc_val", float("nan"))),
+                          acc_tr=float(blob.get("acc_tr", float("nan"))),
+                          epochs_run=0, acc_val_history=[],
+                          epoch_times_s=[], wall_to_acc_s=None)
+        log("    (loaded weights from %s; training skipped)" % cfg.load_model)
+    else:
+        trainer = CbowTrainer(cfg, n_genes, device, ctx, log=log)
+        with timers.phase("train"):
+            res = trainer.train(ps)
     jsonl.emit("train", acc_val=res.acc_val, acc_tr=res.acc_tr,
                stop_epoch=res.stop_epoch, epochs_run=res.epochs_run,
                wall_to_acc088_s=res.wall_to_acc_s)

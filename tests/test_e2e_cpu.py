@@ -146,3 +146,19 @@ def test_config_validation_rejects_bad_values():
                 dict(trainer_path="medium"), dict(kmeans_backend="cuml")):
         with _pt.raises(ValueError):
             G2VecConfig(**{**ok, **bad}).validate()
+
+
+def test_load_model_resumes_without_training(tiny_files, tmp_path):
+    """--save-model then --load-model: steps 5-7 reproduce the same
+    biomarkers without retraining; shape mismatch raises."""
+    ckpt = str(tmp_path / "w.pt")
+    res1 = run(_cfg(tiny_files, tmp_path, save_model=ckpt))
+    out2 = tmp_path / "resume"
+    out2.mkdir()
+    res2 = run(_cfg(tiny_files, out2, load_model=ckpt))
+    assert res2["biomarkers"] == res1["biomarkers"]
+    assert np.allclose(res2["W_ih"], res1["W_ih"])
+    bad = _cfg(tiny_files, tmp_path, load_model=ckpt)
+    bad.hidden = 256
+    with pytest.raises(ValueError, match="load-model"):
+        run(bad)
