@@ -37,3 +37,24 @@ def test_bench_json_contract(tmp_path):
         assert key in cfg, key
     assert cfg["parallelism"] == "dp1"
     assert "synthetic" in d["data"]
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_dp2_contract():
+    """The driver's SCALE launch path: torch.distributed.run with 2 ranks
+    (gloo on CPU) must still produce ONE valid JSON line from rank 0."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29641", "bench.py", "--gpus", "2",
+         "--n-genes", "500", "--n-edges", "6000", "--n-extra", "50",
+         "--n-modules", "6", "--reps", "2", "--len-path", "12",
+         "--steps", "2", "--warmup", "1", "--acc-target-epochs", "2"],
+        capture_output=True, text=True, timeout=500)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines()
+             if l.strip().startswith("{")]
+    assert len(lines) == 1, f"exactly one JSON line expected: {lines}"
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
