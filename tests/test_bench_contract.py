@@ -2,6 +2,7 @@
 protocol and its field set — run the real script as a subprocess on a tiny
 config and validate the schema."""
 import json
+import socket
 import subprocess
 import sys
 
@@ -39,6 +40,12 @@ def test_bench_json_contract(tmp_path):
     assert "synthetic" in d["data"]
 
 
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 @pytest.mark.timeout(600)
 def test_bench_torchrun_dp2_contract():
     """The driver's SCALE launch path: torch.distributed.run with 2 ranks
@@ -46,7 +53,7 @@ def test_bench_torchrun_dp2_contract():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29641", "bench.py", "--gpus", "2",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "2",
          "--n-genes", "500", "--n-edges", "6000", "--n-extra", "50",
          "--n-modules", "6", "--reps", "2", "--len-path", "12",
          "--steps", "2", "--warmup", "1", "--acc-target-epochs", "2"],
