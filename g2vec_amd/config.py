@@ -51,8 +51,14 @@ class G2VecConfig:
                 f"multiple of the 64-lane wavefront in {{64,128,256,512,1024}}")
         if self.dtype not in ("fp32", "bf16", "fp16"):
             raise ValueError(f"dtype must be fp32|bf16|fp16, got {self.dtype}")
+        if self.epochs < 1:
+            raise ValueError("epochs must be >= 1 (the reference always runs "
+                             "at least one epoch, G2Vec.py:262)")
         if self.len_path < 1 or self.len_path > 512:
             raise ValueError("len_path must be in [1, 512] (LDS visited-list budget)")
+        # note: total path nnz (sum of path lengths) is limited to < 2^31
+        # by the int32 kernel index space — enforced at PathSet build time
+        # (paths.py guards integrate_pathsets/subset)
         if self.pcc_mode not in ("auto", "edge", "gemm"):
             raise ValueError(f"bad pcc_mode {self.pcc_mode}")
         if self.trainer_path not in ("fast", "general"):

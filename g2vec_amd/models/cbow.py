@@ -7,8 +7,12 @@ LINEAR, which the MI355X fast path exploits:
   forward:   o_p = sum_{g in p} s_g      with s = W_ih @ W_ho   (one GEMV +
              a scalar segment-sum kernel instead of a [B,G]x[G,h] matmul)
   backward:  dW_ih = c (outer) W_ho,  dW_ho = W_ih^T c,
-             with c = X^T dO in R^G    (rank-1 gradient; the DP all-reduce
-             message is G+h floats, not G*h)
+             with c = X^T dO in R^G    (rank-1 gradient: the per-epoch DP
+             all-reduce message is the G floats of c, not G*h — dW_ho is
+             recomputed per rank from the reduced c, so h floats never
+             travel. A second 2-float metric all-reduce runs per epoch
+             only under early stopping; fixed-epoch runs defer it to one
+             history-sized reduce after the loop.)
 
 The kernel-chain "general" path (gather rows -> wave reduce -> scatter-add,
 SURVEY §2.10 K1-K8) computes the identical math without the collapse and is
@@ -207,7 +211,8 @@ class CbowTrainer:
                 st.dO_buf.copy_(d0)
         return st
 
-    def _epoch_body_fast(self, st, counts_out=None, lrt_slot=None) -> None:
+    def _epoch_body_fast(self, st, counts_out=None, lrt_slot=None,
+                         reduce_counts: bool = True) -> None:
         """One full-batch fast-path epoch as a capturable body: optimizer
         step at W_t, then post-update s + accuracy counts. All inputs and
         outputs live in persistent buffers (s_buf, lrt_buf, counts_buf) so
@@ -215,7 +220,15 @@ class CbowTrainer:
 
         counts_out/lrt_slot override the default buffers — the k-epoch
         block graph records k bodies, each bound to its own slot of a
-        [k,2] counts buffer and a [k] lr_t buffer."""
+        [k,2] counts buffer and a [k] lr_t buffer.
+
+        reduce_counts=False skips the C3 metric all-reduce: fixed-epoch
+        runs consume accuracies only after the loop, so per-epoch counts
+        go into a device-side history that is all-reduced ONCE at the end
+        (identical values — the sum over ranks commutes with the copy into
+        the history). Steady-state collective cost: one grad all-reduce
+        per epoch. Early-stop epochs keep reduce_counts=True (the stop
+        decision reads the global accuracy every epoch)."""
         cfg = self.cfg
         tr, vl = st.tr, st.vl
         if counts_out is None:
@@ -244,7 +257,8 @@ class CbowTrainer:
             ops.cbow_eval_counts_(st.s_buf, st.ev_genes, st.ev_offsets,
                                   st.ev_labels, tr.n_paths, counts_out,
                                   dO=st.dO_buf, inv_b=st.inv_b)
-        self.ctx.allreduce_(counts_out)         # C3: one fused metric reduce
+        if reduce_counts:
+            self.ctx.allreduce_(counts_out)     # C3: 2-float metric reduce
 
     def run_epoch(self, st) -> tuple:
         """One reference epoch: optimizer step(s) at W_t, then post-update
@@ -303,18 +317,19 @@ class CbowTrainer:
         return acc_tr, acc_val
 
     def _ensure_graph(self, st) -> None:
-        """Capture the epoch body into a hipGraph when eligible. Capture
-        only at world==1 for now: a replay-time RCCL issue inside a graph
-        could hang rather than raise, and the eager fallback cannot catch
-        that (collectives are latency-bound at this message size anyway).
-        TODO round 2: validate RCCL+graphs on a real multi-GPU node and
-        lift the gate."""
-        if (self.cfg.use_hipgraph and self.ctx.world == 1 and
-                st.graph is None and not st.graph_failed and
-                st.epoch_idx >= 1):
+        """Capture the epoch body into a hipGraph when eligible. At
+        world>1 the body contains RCCL collectives; capture is gated on a
+        one-time runtime probe (DistContext.graph_capture_ok: capture +
+        replay + verify a tiny all-reduce; G2VEC_DIST_GRAPH=0 opts out).
+        A rank that falls back to eager stays collective-compatible with
+        ranks that replay graphs — both issue the identical collective
+        sequence per epoch."""
+        if (self.cfg.use_hipgraph and st.graph is None and
+                not st.graph_failed and st.epoch_idx >= 1 and
+                self.ctx.graph_capture_ok()):
             try:
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     self._epoch_body_fast(st)
                 st.graph = g            # capture records without executing
             except Exception as e:  # noqa: BLE001
@@ -351,11 +366,16 @@ class CbowTrainer:
         (recording executes nothing and mutates no epoch state). Callable
         ahead of time — bench.py warms it during the untimed warmup so the
         one-time capture never lands in a timed region. Requires at least
-        one prior eager epoch (st.epoch_idx >= 1). Capture is gated to
-        world==1 (same RCCL-inside-graph rationale as _ensure_graph);
-        when gated off the caller's eager tail loop runs every epoch —
-        still with the deferred single readback."""
-        if self.ctx.world != 1 or not self.cfg.use_hipgraph:
+        one prior eager epoch (st.epoch_idx >= 1). At world>1 each
+        recorded body contains the grad all-reduce; capture is gated on
+        the runtime RCCL-capture probe (see _ensure_graph). Bodies are
+        recorded with reduce_counts=False — the kblocked runner all-
+        reduces the accuracy history once after the loop. When capture is
+        unavailable the caller's eager tail loop runs every epoch — still
+        with the deferred readback/reduce."""
+        if self.device.type != "cuda" or not self.cfg.use_hipgraph:
+            return
+        if not self.ctx.graph_capture_ok():
             return
         K = self.KBLOCK
         if getattr(st, "kbufs", None) is None:
@@ -367,10 +387,11 @@ class CbowTrainer:
             klrt, kcounts = st.kbufs[0], st.kbufs[1]
             try:
                 g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     for j in range(K):
                         self._epoch_body_fast(st, counts_out=kcounts[j],
-                                              lrt_slot=klrt[j:j + 1])
+                                              lrt_slot=klrt[j:j + 1],
+                                              reduce_counts=False)
                 st.kgraph = g        # capture records without executing
             except Exception as e:  # noqa: BLE001
                 # capture mutates no epoch state; callers fall back to
@@ -381,9 +402,10 @@ class CbowTrainer:
 
     def kblock_eligible(self, st, n_epochs: int, early_stop: bool) -> bool:
         """Fixed-epoch runs go through the deferred-readback runner: block
-        graphs when world==1 allows capture, eager epoch bodies otherwise
-        (world>1 / --no-hipgraph) — either way zero mid-run D2H."""
-        return (not early_stop and self.device.type == "cuda" and
+        graphs where capture is available, eager epoch bodies otherwise
+        (--no-hipgraph / failed probe / CPU) — either way zero mid-run D2H
+        and ONE metric all-reduce for the whole run."""
+        return (not early_stop and
                 not getattr(st, "kgraph_failed", False) and
                 st.ev_genes is not None and n_epochs >= 2)
 
@@ -391,16 +413,23 @@ class CbowTrainer:
         """Fixed-epoch fast path (early stop OFF): KBLOCK epochs are
         recorded into ONE hipGraph — each recorded body bound to its own
         lr_t slot and [2]-counts slot. Per-epoch counts are appended to a
-        DEVICE-side history with D2D copies and the whole history is read
-        back in a single D2H at the end: zero mid-run host<->device
-        crossings. (A rare per-process driver slow-mode was traced to
-        degraded D2H copies — ~1 ms each regardless of size; with one
-        final read the worst case costs ~1 ms per RUN, not per block.)
+        DEVICE-side history with D2D copies; the whole history is read
+        back in a single D2H at the end AND (world>1) all-reduced in a
+        single collective: zero mid-run host<->device crossings and ONE
+        metric collective per RUN — the steady-state per-epoch collective
+        cost is exactly the grad all-reduce. (A rare per-process driver
+        slow-mode was traced to degraded D2H copies — ~1 ms each
+        regardless of size; with one final read the worst case costs
+        ~1 ms per RUN, not per block.)
         Exact per-epoch semantics: every epoch runs the full optimizer
         pass and both accuracy evals; fixed-epoch runs consume the
-        accuracies only after the loop, so deferring the readback changes
-        nothing observable. Returns the run_epochs_pipelined tuple."""
+        accuracies only after the loop, so deferring the readback and the
+        metric sum-over-ranks (which commutes with the history append)
+        changes nothing observable. Runs on CPU too (eager bodies, same
+        deferred schedule) so the gloo DP tier covers this exact path.
+        Returns the run_epochs_pipelined tuple."""
         K = self.KBLOCK
+        on_gpu = self.device.type == "cuda"
         if st.epoch_idx == 0:    # one eager epoch: warm allocator/state
             warm = [self.run_epoch(st)]
         else:
@@ -419,7 +448,7 @@ class CbowTrainer:
             [ops.tf1_lr_t(self.cfg.lr, self.B1, self.B2, st.t_adam + i)
              for i in range(1, n_rest + 1)],
             dtype=torch.float32, device=self.device)
-        hist_dev = torch.empty(max(n_rest, 1), 2, dtype=torch.float32,
+        hist_dev = torch.zeros(max(n_rest, 1), 2, dtype=torch.float32,
                                device=self.device)
         for b in range(n_blocks):
             klrt.copy_(sched[b * K:(b + 1) * K], non_blocking=True)
@@ -432,11 +461,14 @@ class CbowTrainer:
             st.t_adam += 1
             st.epoch_idx += 1
             self._epoch_body_fast(st, counts_out=st.counts_buf,
-                                  lrt_slot=sched[e:e + 1])
+                                  lrt_slot=sched[e:e + 1],
+                                  reduce_counts=False)
             hist_dev[e].copy_(st.counts_buf, non_blocking=True)
-            if (j + 1) % 64 == 0:       # bound launch-queue depth on long
-                torch.cuda.synchronize()   # eager (world>1) runs
+            if on_gpu and (j + 1) % 64 == 0:   # bound launch-queue depth on
+                torch.cuda.synchronize()       # long eager (world>1) runs
 
+        if n_rest > 0:
+            self.ctx.allreduce_(hist_dev)      # C3: ONE reduce for the run
         cc = hist_dev.cpu()                    # the ONE host readback
         acc_tr = warm[-1][0] if warm else 0.0
         hist = [h[1] for h in warm]

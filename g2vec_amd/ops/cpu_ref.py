@@ -106,7 +106,15 @@ def random_walks(row_ptr: torch.Tensor, col_idx: torch.Tensor, weights: torch.Te
             target = _u01(r) * tot
             c = np.cumsum(wt, dtype=np.float32)
             j = int(np.searchsorted(c, target, side="right"))
-            j = min(j, len(nb) - 1)
+            if j >= len(nb) or wt[j] <= 0.0:
+                # rounding tail: tot (pairwise np.sum) can exceed the
+                # sequential cumsum total in f32, pushing target past
+                # c[-1]. Fall back to the LAST POSITIVE-WEIGHT (unvisited)
+                # neighbor, exactly like the device kernel's ballot
+                # fallback (g2vec_kernels.hip walk_kernel) — never a
+                # visited/zero-weight one (the non-revisit invariant that
+                # paths.py's "paths are sets" assumption relies on).
+                j = int(np.flatnonzero(wt > 0.0)[-1])
             cur = int(nb[j])
         plen = len(visited)
         nodes[wid, :plen] = visited

@@ -330,6 +330,18 @@ torch::Tensor pcc_edges(torch::Tensor zt, torch::Tensor edge_idx,
 
 torch::Tensor corr_gemm(torch::Tensor zt, int64_t n_group) {
   CHECK_DEV(zt); CHECK_CONT(zt); CHECK_F32(zt);
+  // the kernel body is compiled only for gfx950 (MFMA builtins): on any
+  // other arch the launch would be a no-op returning uninitialized C —
+  // trap here instead of silently emitting garbage PCC weights
+  {
+    hipDeviceProp_t prop;
+    int dev = 0;
+    (void)hipGetDevice(&dev);
+    (void)hipGetDeviceProperties(&prop, dev);
+    TORCH_CHECK(std::string(prop.gcnArchName).rfind("gfx950", 0) == 0,
+                "corr_gemm: MFMA kernel is gfx950-only, device reports ",
+                prop.gcnArchName, " (use pcc_mode=edge)");
+  }
   const int G = (int)zt.size(0);
   const int S = (int)zt.size(1);
   const int S4 = ((S + 3) / 4) * 4;

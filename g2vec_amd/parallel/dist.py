@@ -2,11 +2,15 @@
 
 The reference is single-process with zero collectives; the
 framework adds exactly the call sites surveyed there:
-  C1/C2  grad all-reduce (here a single fused bucket: the rank-1 backward
-         collapses dW_ih to c = X^T dO in R^G, so the whole grad message is
-         G+h floats instead of G*h — xGMI ring latency dominates at this
-         size, one bucket is optimal)
-  C3     scalar metric all-reduce (fused into the same bucket)
+  C1/C2  grad all-reduce: the rank-1 backward collapses dW_ih to
+         c = X^T dO in R^G, so the grad message is G floats instead of
+         G*h (dW_ho is recomputed per rank from the reduced c — h floats
+         never travel); one collective, latency-bound at xGMI sizes
+  C3     metric all-reduce (2-float accuracy counts). Early-stop runs
+         issue it per epoch (the stop decision needs it); fixed-epoch
+         runs DEFER it — per-epoch counts accumulate in a device-side
+         history that is all-reduced ONCE after the loop, so the
+         steady-state collective cost is one grad all-reduce per epoch
   C4     initial weight broadcast from rank 0
   C5     all-gather of per-rank walk shards for global dedup
 
@@ -30,6 +34,7 @@ class DistContext:
         self.world = world
         self.device = device
         self.initialized = initialized
+        self._graph_ok: Optional[bool] = None
 
     @property
     def is_primary(self) -> bool:
@@ -72,6 +77,53 @@ class DistContext:
         outs = [torch.zeros_like(padded) for _ in range(self.world)]
         dist.all_gather(outs, padded)
         return [o[:c] for o, c in zip(outs, counts)]
+
+    def graph_capture_ok(self) -> bool:
+        """Can collectives from this process group be captured into a
+        hipGraph? world==1: trivially yes (no collectives execute).
+        world>1 on GPU: RCCL supports stream-capture of its kernels, but
+        rather than assume it, probe ONCE — capture a 2-float all-reduce
+        into a throwaway graph, replay it, and check the math. A capture
+        error falls back cleanly; `G2VEC_DIST_GRAPH=0` opts out entirely.
+        COLLECTIVE: all ranks must call this at the same point (they do —
+        it is only reached from the lockstep epoch loop)."""
+        if self.world == 1:
+            return True
+        if self._graph_ok is None:
+            if (os.environ.get("G2VEC_DIST_GRAPH", "1") == "0"
+                    or self.device.type != "cuda"):
+                self._graph_ok = False
+                return False
+            t = torch.ones(2, dtype=torch.float32, device=self.device)
+            dist.all_reduce(t)          # connect the communicator first:
+            torch.cuda.synchronize()    # capture can't establish channels
+            g = None
+            try:
+                g = torch.cuda.CUDAGraph()
+                # thread_local: RCCL's watchdog/event threads must not
+                # invalidate the capture
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
+                    dist.all_reduce(t)
+            except Exception:  # noqa: BLE001
+                g = None
+            # agree BEFORE replaying: if capture failed on any rank, a
+            # replay elsewhere would launch a collective that rank never
+            # joins (hang, not an exception)
+            flag = torch.tensor([1.0 if g is not None else 0.0],
+                                device=self.device)
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            if float(flag.item()) < 1.0:
+                self._graph_ok = False
+                return False
+            try:
+                g.replay()
+                torch.cuda.synchronize()
+                want = float(self.world) ** 2
+                self._graph_ok = bool(torch.allclose(
+                    t, torch.full_like(t, want)))
+            except Exception:  # noqa: BLE001
+                self._graph_ok = False
+        return self._graph_ok
 
     def shard_range(self, n: int) -> Tuple[int, int]:
         """Contiguous [lo, hi) slice of n items owned by this rank."""

@@ -23,6 +23,18 @@ import torch
 from .walks import WalkSet
 
 
+def _check_i32_nnz(nnz: int) -> None:
+    """The HIP kernels index flat gene instances with int32: a PathSet
+    whose total nnz reaches 2^31 would silently wrap in the .int() casts
+    below (e.g. 1M genes x lenPath 512 x 10 reps x 2 groups). Fail loudly
+    instead."""
+    if nnz >= 2 ** 31:
+        raise OverflowError(
+            f"path set has {nnz} gene instances — the int32 kernel index "
+            f"space holds < 2^31. Reduce len_path/num_repetition or shard "
+            f"walk generation across more ranks.")
+
+
 class PathSet(NamedTuple):
     genes: torch.Tensor     # i32 [nnz]   flat gene indices
     offsets: torch.Tensor   # i32 [P+1]
@@ -80,6 +92,7 @@ def integrate_pathsets(good: WalkSet, poor: WalkSet, n_genes: int
 
     mask = nodes >= 0
     genes = nodes[mask].int()
+    _check_i32_nnz(int(genes.numel()))
     offsets = torch.zeros(len(lengths) + 1, dtype=torch.int64, device=device)
     torch.cumsum(lengths, 0, out=offsets[1:])
 
@@ -104,6 +117,7 @@ def subset(ps: PathSet, idx: torch.Tensor) -> PathSet:
     lens = (offs[1:] - offs[:-1])[idx]
     new_off = torch.zeros(len(idx) + 1, dtype=torch.int64, device=idx.device)
     torch.cumsum(lens, 0, out=new_off[1:])
+    _check_i32_nnz(int(new_off[-1].item()))
     starts = offs[idx]
     # gather flat gene ranges
     seg = torch.repeat_interleave(torch.arange(len(idx), device=idx.device), lens)

@@ -143,9 +143,32 @@ def test_config_validation_rejects_bad_values():
     G2VecConfig(**ok).validate()
     for bad in (dict(hidden=100), dict(dtype="int8"), dict(len_path=0),
                 dict(len_path=1000), dict(pcc_mode="dense"),
-                dict(trainer_path="medium"), dict(kmeans_backend="cuml")):
+                dict(trainer_path="medium"), dict(kmeans_backend="cuml"),
+                dict(epochs=0), dict(epochs=-3)):
         with _pt.raises(ValueError):
             G2VecConfig(**{**ok, **bad}).validate()
+
+
+def test_pathset_nnz_int32_guard():
+    """Total gene instances >= 2^31 must fail loudly before the silent
+    int32 wrap in the offsets cast (paths.py)."""
+    import pytest as _pt
+    import torch
+
+    from g2vec_amd.paths import PathSet, _check_i32_nnz, subset
+
+    _check_i32_nnz(2 ** 31 - 1)          # just under: fine
+    with _pt.raises(OverflowError):
+        _check_i32_nnz(2 ** 31)
+
+    # subset() path: fake offsets whose selected lengths overflow int32
+    big = 2 ** 30
+    ps = PathSet(genes=torch.zeros(4, dtype=torch.int32),
+                 offsets=torch.tensor([0, big, 2 * big, 3 * big],
+                                      dtype=torch.int64),
+                 labels=torch.zeros(3), n_genes=10)
+    with _pt.raises(OverflowError):
+        subset(ps, torch.tensor([0, 1]))
 
 
 def test_load_model_resumes_without_training(tiny_files, tmp_path):
