@@ -25,6 +25,7 @@ Rank 0 prints ONE JSON line on stdout; progress goes to stderr.
 from __future__ import annotations
 
 import argparse
+import dataclasses
 import json
 import sys
 import time
@@ -48,8 +49,14 @@ def log(*a):
 
 
 def build_dataset(seed: int, n_genes: int = 7523, n_edges: int = 298799,
-                  n_extra: int = 2381, n_modules: int = 16):
-    """In-memory ex_*-shaped dataset (one per rank; weak scaling)."""
+                  n_extra: int = 2381, n_modules: int = 16,
+                  shared_frac: float = 0.3, off_frac: float = 0.55):
+    """In-memory ex_*-shaped dataset (one per rank; weak scaling).
+
+    shared_frac/off_frac set the convergence difficulty (calibrated with
+    tools/calibrate_difficulty.py so the seeded val-ACC trajectory climbs
+    gradually to 0.88 over tens of epochs like the published transcript,
+    instead of crossing at epoch 0 on a separable dataset)."""
     rng = np.random.default_rng(seed)
     n_net = n_genes + n_extra
     module = np.full(n_net, -1, dtype=np.int64)
@@ -62,7 +69,8 @@ def build_dataset(seed: int, n_genes: int = 7523, n_edges: int = 298799,
     edge_idx = edge_idx[keep]
     _, labels = synth.synth_clinical(135, 58, seed)
     expr = synth.synth_expression(
-        [f"G{i}" for i in range(n_genes)], labels, module[:n_genes], seed)
+        [f"G{i}" for i in range(n_genes)], labels, module[:n_genes], seed,
+        shared_frac=shared_frac, off_frac=off_frac)
     return expr, np.asarray(labels), edge_idx.astype(np.int32), n_genes
 
 
@@ -82,7 +90,15 @@ def main() -> int:
     ap.add_argument("--n-edges", type=int, default=298799)
     ap.add_argument("--n-extra", type=int, default=2381)
     ap.add_argument("--n-modules", type=int, default=16)
-    ap.add_argument("--acc-target-epochs", type=int, default=60)
+    ap.add_argument("--acc-target-epochs", type=int, default=120)
+    ap.add_argument("--conv-seeds", type=int, default=5,
+                    help="training seeds for the wall-to-ACC>=0.88 probe "
+                         "(same dataset, reshuffled split + fresh init, "
+                         "like the unseeded reference)")
+    ap.add_argument("--real-data", action="store_true",
+                    help="real ex_NETWORK/ex_CLINICAL topology (9,904 "
+                         "genes, 298,799 edges; synthesized expression "
+                         "over the 7,523-gene published intersection)")
     ap.add_argument("--no-hipgraph", action="store_true")
     ap.add_argument("--no-pipeline", action="store_true",
                     help="time the synchronous run_epoch loop instead of "
@@ -106,9 +122,23 @@ def main() -> int:
                       use_hipgraph=not args.no_hipgraph)
 
     # ---- dataset + graphs + walks (input pipeline; measured, not the metric)
-    expr, labels, edge_idx, n_genes = build_dataset(
-        args.seed + 1000 * rank, args.n_genes, args.n_edges, args.n_extra,
-        args.n_modules)
+    if args.real_data:
+        from g2vec_amd.utils import refdata
+        ds = refdata.make_real_dataset(seed=args.seed + 1000 * rank)
+        g2i = {g: i for i, g in enumerate(ds["net_genes"])}
+        keep = np.array([g2i[g] for g in ds["expr_genes"]])
+        idx_of = np.full(len(ds["net_genes"]), -1, np.int64)
+        idx_of[keep] = np.arange(len(keep))
+        e = ds["edge_idx"]
+        m = (idx_of[e[:, 0]] >= 0) & (idx_of[e[:, 1]] >= 0)
+        edge_idx = np.stack([idx_of[e[m, 0]], idx_of[e[m, 1]]],
+                            1).astype(np.int32)
+        expr, labels = ds["expr"], np.asarray(ds["labels"])
+        n_genes = len(ds["expr_genes"])
+    else:
+        expr, labels, edge_idx, n_genes = build_dataset(
+            args.seed + 1000 * rank, args.n_genes, args.n_edges, args.n_extra,
+            args.n_modules)
     expr_t = torch.from_numpy(expr).to(device)
     labels_t = torch.from_numpy(labels).to(device)
     edge_t = torch.from_numpy(edge_idx).to(device)
@@ -127,25 +157,50 @@ def main() -> int:
     log(f"[bench] rank {rank}: {n_walks} walks in {walk_s:.3f}s "
         f"({n_walks / walk_s:.0f} walks/s), {ps.n_paths} paths after dedup")
 
-    # ---- convergence probe: epochs + wall to val-ACC >= 0.88 (untimed work
-    # for the throughput metric, but itself the secondary headline)
-    trainer = CbowTrainer(cfg, n_genes, device, ctx,
-                          log=(lambda *a, **k: None))
-    st = trainer.setup(ps, pre_sharded=True)
-    wall_to_acc = None
+    # ---- convergence probe: epochs + wall to val-ACC >= 0.88, over
+    # --conv-seeds independent training seeds (same dataset; fresh split
+    # shuffle + weight init per seed, mirroring the unseeded reference's
+    # run-to-run variation). Untimed work for the throughput metric, but
+    # itself the secondary headline.
+    conv_runs = []
     acc_val = 0.0
-    conv_t0 = time.perf_counter()
-    for ep in range(args.acc_target_epochs):
-        _acc_tr, acc_val = trainer.run_epoch(st)
-        if acc_val >= 0.88:
-            if on_gpu:
-                torch.cuda.synchronize()
-            wall_to_acc = time.perf_counter() - conv_t0
-            break
-    log(f"[bench] rank {rank}: val-ACC {acc_val:.4f} "
-        f"(wall-to-0.88: {wall_to_acc})")
+    for k in range(max(args.conv_seeds, 1)):
+        cfg_k = dataclasses.replace(cfg, seed=args.seed + 7919 * k)
+        trainer = CbowTrainer(cfg_k, n_genes, device, ctx,
+                              log=(lambda *a, **k2: None))
+        st = trainer.setup(ps, pre_sharded=True)
+        wall_k = None
+        acc_k = 0.0
+        ep_k = None
+        hist_k = []
+        conv_t0 = time.perf_counter()
+        for ep in range(args.acc_target_epochs):
+            _acc_tr, a = trainer.run_epoch(st)
+            hist_k.append(round(a, 4))
+            acc_k = max(acc_k, a)
+            if a >= 0.88:
+                if on_gpu:
+                    torch.cuda.synchronize()
+                wall_k = time.perf_counter() - conv_t0
+                ep_k = ep
+                break
+        conv_runs.append({"seed": cfg_k.seed, "epochs_to_0.88": ep_k,
+                          "wall_s": (round(wall_k, 4) if wall_k else None),
+                          "best_acc": round(acc_k, 4)})
+        if k == 0:
+            acc_val = acc_k
+            wall_to_acc = wall_k
+            log(f"[bench] rank {rank}: seed {cfg_k.seed} trajectory "
+                f"{hist_k[:40]}")
+        log(f"[bench] rank {rank}: conv seed {cfg_k.seed}: best ACC "
+            f"{acc_k:.4f}, 0.88 at epoch {ep_k} ({wall_k} s)")
+    crossed = [r["wall_s"] for r in conv_runs if r["wall_s"] is not None]
+    cross_ep = [r["epochs_to_0.88"] for r in conv_runs
+                if r["epochs_to_0.88"] is not None]
 
     # ---- timed throughput region: fresh state, W warmup + K timed epochs
+    trainer = CbowTrainer(cfg, n_genes, device, ctx,
+                          log=(lambda *a, **k2: None))
     st = trainer.setup(ps, pre_sharded=True)
     n_tr_global = trainer.n_tr_global
     pipelined = (on_gpu and args.trainer_path == "fast"
@@ -190,8 +245,11 @@ def main() -> int:
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_PATHS_PER_SEC, 2),
             "dtype": "fp32",
-            "data": f"synthetic ({args.n_genes} genes/135 samples/"
-                    f"{args.n_edges} network edges; random-init weights)",
+            "data": (f"real ex_NETWORK topology ({n_genes} common genes of "
+                     f"9904/135 real clinical samples; synthetic expression; "
+                     f"random-init weights)" if args.real_data else
+                     f"synthetic ({args.n_genes} genes/135 samples/"
+                     f"{args.n_edges} network edges; random-init weights)"),
             "config": {
                 "model": "g2vec-cbow",
                 "global_batch": n_tr_global,
@@ -208,6 +266,13 @@ def main() -> int:
                 "val_acc": round(acc_val, 4),
                 "wall_to_val_acc_0.88_s": (round(wall_to_acc, 4)
                                            if wall_to_acc else None),
+                "conv_seeds": len(conv_runs),
+                "conv_crossed": len(crossed),
+                "wall_to_0.88_s_median": (round(float(np.median(crossed)), 4)
+                                          if crossed else None),
+                "epochs_to_0.88_median": (float(np.median(cross_ep))
+                                          if cross_ep else None),
+                "conv_runs": conv_runs,
                 "walks_per_sec": round(n_walks / walk_s, 1),
                 "step_includes": "full-batch fwd+bwd+allreduce+dense-Adam "
                                  "+ post-update train/val ACC evals",
