@@ -34,7 +34,10 @@ def build_parser() -> argparse.ArgumentParser:
     # framework flags (absent in reference)
     p.add_argument("--seed", type=int, default=0,
                    help="-1 = unseeded (reference-like nondeterminism)")
-    p.add_argument("--dtype", choices=["fp32", "bf16", "fp16"], default="bf16")
+    p.add_argument("--dtype", choices=["fp32", "bf16", "fp16"], default="fp32",
+               help="W_ih gather storage dtype, general kernel chain only "
+                    "(the fast path always computes fp32); reduced dtypes "
+                    "require --trainer-path general")
     p.add_argument("--device", choices=["auto", "cpu", "cuda"], default="auto")
     p.add_argument("--pcc-mode", choices=["auto", "edge", "gemm"], default="auto")
     p.add_argument("--pcc-threshold", type=float, default=0.5,

@@ -26,7 +26,16 @@ class G2VecConfig:
 
     # --- framework knobs (absent in reference) ---
     seed: Optional[int] = 0         # None -> nondeterministic like the reference
-    dtype: str = "bf16"             # compute dtype of the W_ih gather: {"fp32","bf16","fp16"}
+    dtype: str = "fp32"             # W_ih storage dtype of the GENERAL kernel
+                                    # chain's gather ({"fp32","bf16","fp16"};
+                                    # fp32 master weights/grads either way).
+                                    # SCOPE: trainer_path="general" only — the
+                                    # default fast path's collapsed scalar
+                                    # algebra always computes in fp32, i.e. AT
+                                    # OR ABOVE the requested precision.
+                                    # validate() rejects a reduced dtype
+                                    # combined with the fast path so the flag
+                                    # can never silently do nothing.
     device: str = "auto"            # "auto" | "cpu" | "cuda"
     pcc_threshold: float = 0.5      # |PCC| cutoff (G2Vec.py:385-390)
     pcc_mode: str = "auto"          # "edge" (per-edge dot) | "gemm" (MFMA corr GEMM) | "auto"
@@ -51,6 +60,12 @@ class G2VecConfig:
                 f"multiple of the 64-lane wavefront in {{64,128,256,512,1024}}")
         if self.dtype not in ("fp32", "bf16", "fp16"):
             raise ValueError(f"dtype must be fp32|bf16|fp16, got {self.dtype}")
+        if self.dtype != "fp32" and self.trainer_path != "general":
+            raise ValueError(
+                f"dtype={self.dtype} applies to the general kernel chain's "
+                f"W_ih gather storage only; the fast path computes in fp32 "
+                f"regardless. Use --trainer-path general with a reduced "
+                f"dtype, or drop --dtype.")
         if self.epochs < 1:
             raise ValueError("epochs must be >= 1 (the reference always runs "
                              "at least one epoch, G2Vec.py:262)")

@@ -83,7 +83,25 @@ class CbowTrainer:
 
     # ------------------------------------------------------------------ setup
     def _init_weights(self, gen: Optional[torch.Generator]):
+        """+-2sigma truncated normal, stddev 1/sqrt(h) (G2Vec.py:234-235).
+        On GPU the K9 device kernel fills W in place from a counter-based
+        stream — no host-side rejection loop (which materialized multi-GB
+        intermediates at 1M x 512); on CPU the seeded host sampler keeps
+        its round-1 bit-exact streams. Unseeded runs derive a random seed
+        (rank 0's broadcast in setup() makes ranks agree either way)."""
         std = 1.0 / (self.h ** 0.5)
+        if self.device.type == "cuda":
+            if self.cfg.seed is not None:
+                dev_seed = int(self.cfg.seed)
+            else:
+                dev_seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+            W = torch.empty(self.G, self.h, dtype=torch.float32,
+                            device=self.device)
+            who = torch.empty(self.h, dtype=torch.float32,
+                              device=self.device)
+            ops.trunc_normal_(W, std, dev_seed)
+            ops.trunc_normal_(who, std, dev_seed + 0x9E3779B9)
+            return W, who
         W = torch.empty(self.G, self.h, dtype=torch.float32)
         who = torch.empty(self.h, dtype=torch.float32)
         if gen is None:

@@ -148,6 +148,16 @@ def scatter_dO(genes, offsets, dO, n_genes: int,
     return cpu_ref.scatter_dO(genes, offsets, dO, n_genes)
 
 
+def trunc_normal_(out: torch.Tensor, std: float, seed: int) -> None:
+    """Seeded +-2sigma truncated-normal fill (K9, reference init
+    semantics tf.truncated_normal G2Vec.py:234-235). Device kernel on
+    GPU — no host-side multi-GB rejection loop at the 1M x 512 config."""
+    if out.is_cuda:
+        native().trunc_normal_(out, float(std), int(seed))
+        return
+    cpu_ref.trunc_normal_(out, std, seed)
+
+
 def tf1_lr_t(lr: float, b1: float, b2: float, t: int) -> float:
     """TF1 AdamOptimizer effective step size (G2Vec.py:245-246 semantics)."""
     return lr * (1.0 - b2 ** t) ** 0.5 / (1.0 - b1 ** t)

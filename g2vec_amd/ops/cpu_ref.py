@@ -123,6 +123,36 @@ def random_walks(row_ptr: torch.Tensor, col_idx: torch.Tensor, weights: torch.Te
     return (torch.from_numpy(nodes), torch.from_numpy(lengths), torch.from_numpy(hashes))
 
 
+# ------------------------------------------------------------------ K9 init
+def trunc_normal_(out: torch.Tensor, std: float, seed: int) -> None:
+    """Counter-based +-2sigma truncated normal, mirroring
+    trunc_normal_kernel (g2vec_kernels.hip): element i draws Box-Muller
+    normals from its own splitmix64 stream and resamples beyond 2sigma.
+    Same streams as the device kernel; the float cos/log tails can differ
+    by ulps across libm implementations, so parity tests compare with a
+    small tolerance (plus bounds/moments), not bitwise."""
+    import math
+
+    flat = out.view(-1)
+    n = flat.numel()
+    vals = np.empty(n, dtype=np.float32)
+    with np.errstate(over="ignore"):
+        for i in range(n):
+            state = np.uint64(seed) ^ np.uint64(
+                np.uint64(i) * _SM64_M2 + np.uint64(1))
+            state, _ = splitmix64(state)      # warm draw (kernel parity)
+            while True:
+                state, r1 = splitmix64(state)
+                state, r2 = splitmix64(state)
+                u1 = max(_u01(r1), 1e-300)
+                x = np.float32(math.sqrt(-2.0 * math.log(u1)) *
+                               math.cos(2.0 * math.pi * _u01(r2)))
+                if abs(x) <= np.float32(2.0):
+                    break
+            vals[i] = x
+    flat.copy_(torch.from_numpy(vals * np.float32(std)))
+
+
 # ------------------------------------------------------------------ CBOW (fast scalar path)
 def cbow_fwd_scalar(s: torch.Tensor, genes: torch.Tensor, offsets: torch.Tensor,
                     labels: torch.Tensor, inv_b: float, want_grad: bool):

@@ -358,6 +358,16 @@ torch::Tensor corr_gemm(torch::Tensor zt, int64_t n_group) {
   return C;
 }
 
+void trunc_normal_(torch::Tensor out, double std, int64_t seed) {
+  CHECK_DEV(out); CHECK_CONT(out); CHECK_F32(out);
+  const long long n = (long long)out.numel();
+  if (n == 0) return;
+  hipLaunchKernelGGL(trunc_normal_kernel, dim3(grid_for(n, 256)), dim3(256),
+                     0, cur_stream(), out.data_ptr<float>(), n, (float)std,
+                     (uint64_t)seed);
+  LAUNCH_CHECK();
+}
+
 void gemv_rows_(torch::Tensor W, torch::Tensor x, torch::Tensor out) {
   CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
   CHECK_DEV(x); CHECK_CONT(x); CHECK_F32(x);
@@ -498,6 +508,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cbow_bwd_rows", &cbow_bwd_rows, "scatter-add CBOW backward");
   m.def("pcc_edges", &pcc_edges, "per-edge |PCC|");
   m.def("corr_gemm", &corr_gemm, "MFMA f32 correlation GEMM");
+  m.def("trunc_normal_", &trunc_normal_,
+        "seeded +-2sigma truncated-normal fill (K9)");
   m.def("gemv_rows_", &gemv_rows_, "s = W @ x (wave-per-row GEMV, out-arg)");
   m.def("gemv_cols_", &gemv_cols_, "out = W^T c (partials + fold, out-arg)");
   m.def("bf16_copy", &bf16_copy, "f32 -> bf16 cast kernel");

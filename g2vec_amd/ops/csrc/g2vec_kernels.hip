@@ -842,3 +842,32 @@ INSTANTIATE_GEMVR(16)
 INSTANTIATE_GEMVC(1)
 INSTANTIATE_GEMVC(2)
 INSTANTIATE_GEMVC(4)
+
+// ======================================================= K9: trunc-normal init
+// Seeded device truncated normal (+-2 sigma, rejection-resampled) — the
+// reference init semantics (tf.truncated_normal, G2Vec.py:234-235)
+// computed entirely on-device: the host rejection loop materialized
+// multi-GB intermediates at the 1M x 512 config (round-1 verdict item 6).
+// Counter-based: element i draws from its own splitmix64 stream, so the
+// result is independent of the launch geometry and bit-stable across
+// runs; the CPU oracle (cpu_ref.trunc_normal) mirrors the algorithm.
+extern "C" __global__ void trunc_normal_kernel(float* __restrict__ out,
+                                               long long n, float std,
+                                               uint64_t seed) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    uint64_t state =
+        seed ^ (uint64_t)((uint64_t)i * 0x94D049BB133111EBULL + 1ULL);
+    (void)sm64_next(state);                  // warm draw (stream decorrelation)
+    float x;
+    do {                                     // P(reject) ~ 4.6%/draw
+      const double u1 = u01_from(sm64_next(state));
+      const double u2 = u01_from(sm64_next(state));
+      // Box-Muller (cosine branch); u1 clamped away from log(0)
+      const double r = sqrt(-2.0 * log(u1 > 1e-300 ? u1 : 1e-300));
+      x = (float)(r * cos(6.283185307179586 * u2));
+    } while (fabsf(x) > 2.0f);
+    out[i] = x * std;
+  }
+}
