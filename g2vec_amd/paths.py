@@ -46,59 +46,77 @@ class PathSet(NamedTuple):
         return int(self.labels.shape[0])
 
 
-def _unique_first(hashes: torch.Tensor) -> torch.Tensor:
-    """Indices of one representative per distinct hash (deterministic:
-    the representative with the smallest original index)."""
+def _unique_first(hashes: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(indices of one representative per distinct hash, their SORTED hash
+    values). Deterministic: the representative with the smallest original
+    index (stable sort keeps original order within equal hashes)."""
     sh, perm = torch.sort(hashes, stable=True)
     first = torch.ones_like(sh, dtype=torch.bool)
     if sh.numel() > 1:
         first[1:] = sh[1:] != sh[:-1]
-    # within equal hashes stable sort keeps original order -> first is min-index
-    return perm[first]
+    return perm[first], sh[first]
+
+
+def _isin_sorted(a_sorted: torch.Tensor, b_sorted: torch.Tensor) -> torch.Tensor:
+    """bool mask: a_sorted[i] present in b_sorted. Both inputs sorted —
+    one binary search instead of torch.isin's internal sort."""
+    if b_sorted.numel() == 0:
+        return torch.zeros_like(a_sorted, dtype=torch.bool)
+    pos = torch.searchsorted(b_sorted, a_sorted)
+    pos = pos.clamp_(max=b_sorted.numel() - 1)
+    return b_sorted[pos] == a_sorted
 
 
 def integrate_pathsets(good: WalkSet, poor: WalkSet, n_genes: int
                        ) -> Tuple[PathSet, torch.Tensor, int]:
-    """Returns (pathset, gene_freq i64 [G] with values {0,1,2}, n_genes_in_paths)."""
-    device = good.nodes.device
-    kept = []
-    kept_hashes = []
-    for ws in (good, poor):
-        idx = _unique_first(ws.hashes)
-        kept.append(idx)
-        kept_hashes.append(ws.hashes[idx])
-    common_g = torch.isin(kept_hashes[0], kept_hashes[1])
-    common_p = torch.isin(kept_hashes[1], kept_hashes[0])
-    keep_g = kept[0][~common_g]
-    keep_p = kept[1][~common_p]
+    """Returns (pathset, gene_freq i64 [G] with values {0,1,2}, n_genes_in_paths).
 
-    parts_nodes, parts_len, parts_lab = [], [], []
-    for ws, keep, lab in ((good, keep_g, 0.0), (poor, keep_p, 1.0)):
-        if keep.numel() == 0:
-            continue
-        parts_nodes.append(ws.nodes[keep])
-        parts_len.append(ws.lengths[keep])
-        parts_lab.append(torch.full((keep.numel(),), lab, dtype=torch.float32,
-                                    device=device))
-    if not parts_nodes:
+    Device-friendly schedule (step-3 hot path — this function was 23 ms
+    of the 28 ms warmed step-3 wall in round 1): per-group dedup keeps
+    its hashes sorted so cross-group common-path removal is a binary
+    search (no second sort), and the kept walks' gene lists are packed
+    into CSR with a flat gather over exactly the kept instances instead
+    of a boolean mask over the whole padded [n_walks, len_path] buffer
+    (12M elements at ex_* scale)."""
+    device = good.nodes.device
+    kept, kept_sorted_h = [], []
+    for ws in (good, poor):
+        idx, sh = _unique_first(ws.hashes)
+        kept.append(idx)
+        kept_sorted_h.append(sh)
+    keep_g = kept[0][~_isin_sorted(kept_sorted_h[0], kept_sorted_h[1])]
+    keep_p = kept[1][~_isin_sorted(kept_sorted_h[1], kept_sorted_h[0])]
+    n_g, n_p = int(keep_g.numel()), int(keep_p.numel())
+    P = n_g + n_p
+    if P == 0:
         empty = torch.zeros(0, dtype=torch.int32, device=device)
         return (PathSet(empty, torch.zeros(1, dtype=torch.int32, device=device),
                         torch.zeros(0, dtype=torch.float32, device=device), n_genes),
                 torch.full((n_genes,), 2, dtype=torch.int64, device=device), 0)
 
-    nodes = torch.cat(parts_nodes)           # [P, L]
-    lengths = torch.cat(parts_len).long()    # [P]
-    labels = torch.cat(parts_lab)
+    if good.nodes.shape[1] != poor.nodes.shape[1]:
+        raise ValueError("good/poor walk sets must share len_path")
+    L = int(good.nodes.shape[1])
+    lengths = torch.cat([good.lengths[keep_g], poor.lengths[keep_p]]).long()
+    labels = torch.zeros(P, dtype=torch.float32, device=device)
+    labels[n_g:] = 1.0
 
-    mask = nodes >= 0
-    genes = nodes[mask].int()
-    _check_i32_nnz(int(genes.numel()))
-    offsets = torch.zeros(len(lengths) + 1, dtype=torch.int64, device=device)
+    offsets = torch.zeros(P + 1, dtype=torch.int64, device=device)
     torch.cumsum(lengths, 0, out=offsets[1:])
+    nnz = int(offsets[-1].item())
+    _check_i32_nnz(nnz)
 
-    # gene frequencies (paths are sets: non-revisiting walks never repeat a gene)
-    seg = torch.repeat_interleave(torch.arange(len(lengths), device=device), lengths)
-    is_poor = labels[seg] > 0.5
+    # flat gather of exactly the kept instances: row starts in the padded
+    # buffers, one arange re-based per path
+    seg = torch.repeat_interleave(torch.arange(P, device=device), lengths)
+    pos = torch.arange(nnz, device=device) - offsets[seg]
+    row_start = torch.cat([keep_g, keep_p + good.nodes.shape[0]]) * L
+    flat = torch.cat([good.nodes.reshape(-1), poor.nodes.reshape(-1)])
+    genes = flat[row_start[seg] + pos].int()
+
+    # gene frequencies (paths are sets: non-revisiting walks never repeat
+    # a gene, so per-path counts are 0/1)
+    is_poor = seg >= n_g
     cnt_g = torch.bincount(genes[~is_poor].long(), minlength=n_genes)
     cnt_p = torch.bincount(genes[is_poor].long(), minlength=n_genes)
     freq = torch.full((n_genes,), 2, dtype=torch.int64, device=device)
