@@ -67,8 +67,19 @@ def _isin_sorted(a_sorted: torch.Tensor, b_sorted: torch.Tensor) -> torch.Tensor
     return b_sorted[pos] == a_sorted
 
 
-def integrate_pathsets(good: WalkSet, poor: WalkSet, n_genes: int
-                       ) -> Tuple[PathSet, torch.Tensor, int]:
+class _NullTimer:
+    def __call__(self, name):
+        return self
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
+
+
+def integrate_pathsets(good: WalkSet, poor: WalkSet, n_genes: int,
+                       timers=None) -> Tuple[PathSet, torch.Tensor, int]:
     """Returns (pathset, gene_freq i64 [G] with values {0,1,2}, n_genes_in_paths).
 
     Device-friendly schedule (step-3 hot path — this function was 23 ms
@@ -78,15 +89,18 @@ def integrate_pathsets(good: WalkSet, poor: WalkSet, n_genes: int
     into CSR with a flat gather over exactly the kept instances instead
     of a boolean mask over the whole padded [n_walks, len_path] buffer
     (12M elements at ex_* scale)."""
+    t = timers or _NullTimer()
     device = good.nodes.device
     kept, kept_sorted_h = [], []
-    for ws in (good, poor):
-        idx, sh = _unique_first(ws.hashes)
-        kept.append(idx)
-        kept_sorted_h.append(sh)
-    keep_g = kept[0][~_isin_sorted(kept_sorted_h[0], kept_sorted_h[1])]
-    keep_p = kept[1][~_isin_sorted(kept_sorted_h[1], kept_sorted_h[0])]
-    n_g, n_p = int(keep_g.numel()), int(keep_p.numel())
+    with t("integ.dedup_sort"):
+        for ws in (good, poor):
+            idx, sh = _unique_first(ws.hashes)
+            kept.append(idx)
+            kept_sorted_h.append(sh)
+    with t("integ.common_search"):
+        keep_g = kept[0][~_isin_sorted(kept_sorted_h[0], kept_sorted_h[1])]
+        keep_p = kept[1][~_isin_sorted(kept_sorted_h[1], kept_sorted_h[0])]
+        n_g, n_p = int(keep_g.numel()), int(keep_p.numel())
     P = n_g + n_p
     if P == 0:
         empty = torch.zeros(0, dtype=torch.int32, device=device)
@@ -97,32 +111,35 @@ def integrate_pathsets(good: WalkSet, poor: WalkSet, n_genes: int
     if good.nodes.shape[1] != poor.nodes.shape[1]:
         raise ValueError("good/poor walk sets must share len_path")
     L = int(good.nodes.shape[1])
-    lengths = torch.cat([good.lengths[keep_g], poor.lengths[keep_p]]).long()
-    labels = torch.zeros(P, dtype=torch.float32, device=device)
-    labels[n_g:] = 1.0
-
-    offsets = torch.zeros(P + 1, dtype=torch.int64, device=device)
-    torch.cumsum(lengths, 0, out=offsets[1:])
-    nnz = int(offsets[-1].item())
-    _check_i32_nnz(nnz)
+    with t("integ.offsets"):
+        lengths = torch.cat([good.lengths[keep_g], poor.lengths[keep_p]]).long()
+        labels = torch.zeros(P, dtype=torch.float32, device=device)
+        labels[n_g:] = 1.0
+        offsets = torch.zeros(P + 1, dtype=torch.int64, device=device)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        nnz = int(offsets[-1].item())
+        _check_i32_nnz(nnz)
 
     # flat gather of exactly the kept instances: row starts in the padded
     # buffers, one arange re-based per path
-    seg = torch.repeat_interleave(torch.arange(P, device=device), lengths)
-    pos = torch.arange(nnz, device=device) - offsets[seg]
-    row_start = torch.cat([keep_g, keep_p + good.nodes.shape[0]]) * L
-    flat = torch.cat([good.nodes.reshape(-1), poor.nodes.reshape(-1)])
-    genes = flat[row_start[seg] + pos].int()
+    with t("integ.seg"):
+        seg = torch.repeat_interleave(torch.arange(P, device=device), lengths)
+        pos = torch.arange(nnz, device=device) - offsets[seg]
+    with t("integ.gather"):
+        row_start = torch.cat([keep_g, keep_p + good.nodes.shape[0]]) * L
+        flat = torch.cat([good.nodes.reshape(-1), poor.nodes.reshape(-1)])
+        genes = flat[row_start[seg] + pos].int()
 
     # gene frequencies (paths are sets: non-revisiting walks never repeat
     # a gene, so per-path counts are 0/1)
-    is_poor = seg >= n_g
-    cnt_g = torch.bincount(genes[~is_poor].long(), minlength=n_genes)
-    cnt_p = torch.bincount(genes[is_poor].long(), minlength=n_genes)
-    freq = torch.full((n_genes,), 2, dtype=torch.int64, device=device)
-    freq[cnt_g > cnt_p] = 0
-    freq[cnt_p > cnt_g] = 1
-    n_in_paths = int(((cnt_g + cnt_p) > 0).sum().item())
+    with t("integ.freq"):
+        is_poor = seg >= n_g
+        cnt_g = torch.bincount(genes[~is_poor].long(), minlength=n_genes)
+        cnt_p = torch.bincount(genes[is_poor].long(), minlength=n_genes)
+        freq = torch.full((n_genes,), 2, dtype=torch.int64, device=device)
+        freq[cnt_g > cnt_p] = 0
+        freq[cnt_p > cnt_g] = 1
+        n_in_paths = int(((cnt_g + cnt_p) > 0).sum().item())
 
     ps = PathSet(genes.contiguous(), offsets.int().contiguous(),
                  labels.contiguous(), n_genes)

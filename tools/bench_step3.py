@@ -18,7 +18,8 @@ import time
 import numpy as np
 import torch
 
-sys.path.insert(0, ".")
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from bench import build_dataset                      # noqa: E402
 from g2vec_amd.graph import (build_group_graph, edge_pcc_weights,  # noqa: E402
@@ -28,24 +29,28 @@ from g2vec_amd.walks import generate_walks           # noqa: E402
 
 
 class T:
+    """Sync-bracketed phase timer; reentrant (nested phases nest on a
+    stack, so integrate_pathsets' sub-timers can share the instance)."""
+
     def __init__(self, dev):
         self.dev = dev
         self.acc = {}
+        self.stack = []
 
     def __call__(self, name):
-        self.name = name
+        self.pending = name
         return self
 
     def __enter__(self):
         if self.dev.type == "cuda":
             torch.cuda.synchronize()
-        self.t0 = time.perf_counter()
+        self.stack.append((self.pending, time.perf_counter()))
 
     def __exit__(self, *a):
         if self.dev.type == "cuda":
             torch.cuda.synchronize()
-        self.acc[self.name] = self.acc.get(self.name, 0.0) + (
-            time.perf_counter() - self.t0)
+        name, t0 = self.stack.pop()
+        self.acc[name] = self.acc.get(name, 0.0) + (time.perf_counter() - t0)
 
 
 def main():
@@ -119,7 +124,7 @@ def main():
             walksets.append(ws)
         with t("integrate"):
             ps, _freq, _nip = integrate_pathsets(walksets[0], walksets[1],
-                                                 n_genes)
+                                                 n_genes, timers=t)
         if dev.type == "cuda":
             torch.cuda.synchronize()
         wall += time.perf_counter() - w0

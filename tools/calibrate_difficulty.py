@@ -20,7 +20,8 @@ import time
 import numpy as np
 import torch
 
-sys.path.insert(0, ".")
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from bench import build_dataset                      # noqa: E402
 from g2vec_amd.config import G2VecConfig             # noqa: E402
