@@ -55,5 +55,18 @@ def main():
     bench("item() sync", lambda: int(lens[-1].item()))
 
 
+def skew():
+    """Hub-skewed gene histogram: atomic contention check."""
+    dev = torch.device("cuda")
+    G = 7523
+    nnz = 1_500_000
+    g = (torch.randn(nnz, device=dev).abs() * (G / 20)).long().clamp_(max=G - 1)
+    bench("bincount SKEWED int64", lambda: torch.bincount(g, minlength=G))
+    onesi = torch.ones(nnz, dtype=torch.int32, device=dev)
+    bench("index_add SKEWED i32", lambda: torch.zeros(
+        G, dtype=torch.int32, device=dev).index_add_(0, g, onesi))
+
+
 if __name__ == "__main__":
     main()
+    skew()
