@@ -86,12 +86,17 @@ def main():
     # its wall is reported separately as step3_cold_ms
     if dev.type == "cuda":
         torch.cuda.synchronize()
+    tc = T(dev)
     c0 = time.perf_counter()
     wsets = []
     for group in (0, 1):
-        g = build_group_graph(expr_t, labels_t, group, edge_t, n_genes)
-        wsets.append(generate_walks(g, args.len_path, args.reps, 999, group))
-    integrate_pathsets(wsets[0], wsets[1], n_genes)
+        with tc("cold.graph"):
+            g = build_group_graph(expr_t, labels_t, group, edge_t, n_genes)
+        with tc("cold.walks"):
+            wsets.append(generate_walks(g, args.len_path, args.reps, 999,
+                                        group))
+    with tc("cold.integrate"):
+        integrate_pathsets(wsets[0], wsets[1], n_genes)
     if dev.type == "cuda":
         torch.cuda.synchronize()
     cold_s = time.perf_counter() - c0
@@ -145,6 +150,8 @@ def main():
         "n_genes": n_genes, "n_walks": n_walks, "n_paths": n_paths,
         "real_data": bool(args.real_data),
         "step3_cold_ms": round(cold_s * 1e3, 3),
+        "cold_phases_ms": {k: round(v * 1e3, 3)
+                           for k, v in sorted(tc.acc.items())},
         "step3_wall_ms": round(wall / args.iters * 1e3, 3),
         "walks_per_sec": round(n_walks * args.iters / wall, 1),
         "phases_ms": {k: round(v / args.iters * 1e3, 3)
