@@ -51,6 +51,11 @@ def build_parser() -> argparse.ArgumentParser:
                    help="reproduce the shipped L-group disambiguation bug "
                         "(G2Vec.py:186-194, SURVEY 2.9)")
     p.add_argument("--no-early-stop", action="store_true")
+    p.add_argument("--earlystop-every", type=int, default=1,
+                   help="read early-stop accuracies every K epochs and "
+                        "deterministically replay to the dip on stop "
+                        "(same trajectory/stop/weights; 1/K the "
+                        "collectives and readbacks; GPU fast path)")
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph capture of the training epoch")
     p.add_argument("--save-paths", type=str, default="")
@@ -75,7 +80,8 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         pcc_mode=a.pcc_mode, pcc_threshold=a.pcc_threshold,
         kmeans_backend=a.kmeans, trainer_path=a.trainer_path,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
-        early_stop=not a.no_early_stop, save_paths=a.save_paths,
+        early_stop=not a.no_early_stop, earlystop_every=a.earlystop_every,
+        save_paths=a.save_paths,
         load_paths=a.load_paths, save_model=a.save_model,
         load_model=a.load_model,
         log_jsonl=a.log_jsonl,

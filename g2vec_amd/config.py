@@ -42,6 +42,13 @@ class G2VecConfig:
     kmeans_backend: str = "auto"    # "sklearn" (reference parity) | "torch" (scales) | "auto"
     compat_lgroup_bug: bool = False  # reproduce the shipped G2Vec.py:186-194 behaviour
     early_stop: bool = True
+    earlystop_every: int = 1        # early-stop accuracy readback granularity:
+                                    # 1 = per-epoch (reference semantics and
+                                    # schedule); k>1 = read accuracies every k
+                                    # epochs and deterministically REPLAY to
+                                    # the dip on stop — same trajectory/stop/
+                                    # weights, 1/k the collectives + D2H
+                                    # (opt-in; see cbow.run_epochs_kgranular)
     batch_size: int = 0             # 0 = full batch (reference semantics, G2Vec.py:264)
     trainer_path: str = "fast"      # "fast": collapsed rank-1 path (linear-net algebra)
                                     # "general": full gather/scatter kernel chain (K1-K8)
@@ -66,6 +73,8 @@ class G2VecConfig:
                 f"W_ih gather storage only; the fast path computes in fp32 "
                 f"regardless. Use --trainer-path general with a reduced "
                 f"dtype, or drop --dtype.")
+        if self.earlystop_every < 1:
+            raise ValueError("earlystop_every must be >= 1")
         if self.epochs < 1:
             raise ValueError("epochs must be >= 1 (the reference always runs "
                              "at least one epoch, G2Vec.py:262)")

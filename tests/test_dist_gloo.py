@@ -245,3 +245,53 @@ def test_dp2_pipelined_epochs_match_sync():
         p.join(timeout=120)
         assert p.exitcode == 0
     assert len(hist_dp) >= 2 and np.isfinite(W_dp).all()
+
+
+def _kgranular_worker(rank, world, port, out):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = _CountingCtx(rank, world, torch.device("cpu"))
+        ps = _pathset(G=60, P=260, seed=21)
+        cfg = G2VecConfig(hidden=64, epochs=30, early_stop=True, seed=6,
+                          device="cpu", dtype="fp32")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                         log=lambda *a, **k: None)
+        res_sync = tr.train(ps)            # CPU train() = sync loop
+        tr2 = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                          log=lambda *a, **k: None)
+        st = tr2.setup(ps)
+        n0 = ctx.n_allreduce
+        hist, stop, W, _who, _ = tr2.run_epochs_kgranular(st, cfg.epochs, 8)
+        n_ar = ctx.n_allreduce - n0
+        # exact parity with the per-epoch sync loop, on every rank
+        assert stop == res_sync.stop_epoch
+        assert hist == pytest.approx(res_sync.acc_val_history, abs=0)
+        assert torch.allclose(W, res_sync.W_ih, atol=0)
+        if rank == 0:
+            out.put((stop, len(hist), n_ar))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_kgranular_early_stop_parity_and_schedule():
+    """k-granular early stop under world_size=2: bitwise the sync
+    trajectory/stop/weights, with one metric all-reduce per 8-epoch
+    block instead of one per epoch."""
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    out = ctxm.Queue()
+    procs = [ctxm.Process(target=_kgranular_worker, args=(r, 2, port, out))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    stop, n_hist, n_ar = out.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert stop >= 0 or n_hist == 30
+    # schedule: warm epoch (grad+counts) + per-epoch grads + one metric
+    # reduce per block (+ replay grads on stop) — strictly fewer than
+    # the 2-per-epoch per-epoch schedule
+    assert n_ar < 2 * n_hist

@@ -168,3 +168,43 @@ def test_pipelined_epochs_match_sync_loop():
         assert torch.allclose(W, res_sync.W_ih, atol=1e-6)
         if early_stop and stop >= 0:
             assert hist[stop + 1] < hist[stop]
+
+
+@pytest.mark.parametrize("k", [2, 3, 8])
+def test_kgranular_early_stop_matches_sync(k):
+    """run_epochs_kgranular (accuracy readback every k epochs +
+    deterministic replay on the dip) must reproduce the sync loop's
+    trajectory, stop epoch, and final weights BITWISE — the epoch body
+    has no RNG, so the replay is exact (round-1 verdict item 8)."""
+    ps = _random_pathset(G=60, P=240, seed=9)
+    cfg = G2VecConfig(hidden=64, epochs=40, early_stop=True, seed=2,
+                      device="cpu", dtype="fp32")
+    ref = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                      log=lambda *a, **kw: None)
+    res = ref.train(ps)
+    assert res.stop_epoch >= 0, "fixture must actually early-stop"
+
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **kw: None)
+    st = tr.setup(ps)
+    hist, stop, W, who, _ = tr.run_epochs_kgranular(st, cfg.epochs, k)
+    assert stop == res.stop_epoch
+    assert hist == pytest.approx(res.acc_val_history, abs=0)
+    assert torch.equal(W, res.W_ih)
+
+
+def test_kgranular_no_dip_runs_all_epochs():
+    ps = _random_pathset(G=40, P=120, seed=3)
+    cfg = G2VecConfig(hidden=64, epochs=6, early_stop=False, seed=1,
+                      device="cpu", dtype="fp32")
+    ref = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                      log=lambda *a, **kw: None)
+    res = ref.train(ps)
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **kw: None)
+    st = tr.setup(ps)
+    hist, stop, W, _who, _ = tr.run_epochs_kgranular(st, 6, 4)
+    # no dip consumed: every epoch's accuracy equals the sync loop's
+    if stop == -1:
+        assert hist == pytest.approx(res.acc_val_history, abs=0)
+        assert torch.equal(W, res.W_ih)
