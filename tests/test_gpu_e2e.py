@@ -143,3 +143,35 @@ def test_kblock_epochs_match_sync_loop(tmp_path):
     assert stop == -1 and len(hist_k) == N_EP
     assert hist_k == pytest.approx(hist_sync, abs=1e-6)
     assert torch.allclose(W, st1.W, atol=1e-6)
+
+
+@pytest.mark.timeout(600)
+def test_kgranular_early_stop_matches_per_epoch_on_gpu():
+    """--earlystop-every on GPU (hipGraph blocks + snapshot/replay) must
+    reproduce the per-epoch pipelined runner's trajectory, stop epoch,
+    and weights bitwise."""
+    import numpy as np
+    from g2vec_amd.models.cbow import CbowTrainer
+    from g2vec_amd.paths import PathSet
+    rng = np.random.default_rng(23)
+    G, P = 250, 500
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 14))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda", 0)
+    ps = PathSet(torch.tensor(genes, dtype=torch.int32, device=dev),
+                 torch.tensor(offs, dtype=torch.int32, device=dev),
+                 torch.tensor(labels, device=dev), G)
+    cfg = G2VecConfig(hidden=128, epochs=40, early_stop=True, seed=8,
+                      device="cuda", dtype="fp32")
+    ref = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None)
+    res = ref.train(ps)                       # per-epoch pipelined runner
+    tr = CbowTrainer(cfg, G, dev, log=lambda *a, **k: None)
+    st = tr.setup(ps)
+    hist, stop, W, _who, _ = tr.run_epochs_kgranular(st, cfg.epochs, 8)
+    assert stop == res.stop_epoch
+    assert hist == pytest.approx(res.acc_val_history, abs=0)
+    assert torch.equal(W.cpu(), res.W_ih.cpu())
