@@ -98,10 +98,20 @@ def _label_propagation(n: int, u: np.ndarray, v: np.ndarray, k: int,
 
 def make_real_dataset(seed: int = 0, n_common: int = N_COMMON_PUBLISHED,
                       n_modules: int = 32, lp_iters: int = 4,
-                      shared_frac: float = 0.15, off_frac: float = 0.45,
-                      min_module: int = 30,
+                      shared_frac: float = 0.1, off_frac: float = 0.45,
+                      min_module: int = 30, dead_frac: float = 0.0,
                       ref_dir: Optional[str] = None) -> Dict:
     """Real network + real clinical + synthesized expression.
+
+    Difficulty note: with the default knobs (calibrated sweep in
+    tools/calibrate_difficulty.py's real-topology variant) the seeded
+    val-ACC climbs ~0.58 -> ~0.83 and plateaus BELOW the published
+    0.8837 — every gene is walked from in both group graphs, and on the
+    real hub-dense topology the off-class walks over a module's genes
+    are genuinely label-ambiguous for the linear model. The real-data
+    variant's job is the real TOPOLOGY (hub rows >256, real loaders,
+    count invariants); the published-difficulty convergence headline
+    lives on the calibrated synthetic ex_* config (bench.py defaults).
 
     Returns {'expr' f32 [S, n_common] (samples x chosen genes),
     'expr_genes', 'samples', 'labels', 'net_genes', 'edge_idx' (full real
@@ -125,6 +135,28 @@ def make_real_dataset(seed: int = 0, n_common: int = N_COMMON_PUBLISHED,
     module = _label_propagation(n_common, u, v, n_modules, lp_iters, seed)
     sizes = np.bincount(module, minlength=n_modules)
     module = np.where(sizes[module] >= min_module, module, -1)
+    # renumber communities by ASCENDING size: module_activity marks the
+    # first round(K*shared_frac) ids shared-in-both-classes and
+    # alternates good/poor over the rest, so with ascending ids the
+    # ambiguous (shared) modules are the SMALLEST communities and the
+    # large ones split between the classes in size-interleaved order —
+    # otherwise the giant label-prop community can land on a shared id
+    # and erase a third of the label signal
+    live_ids = np.unique(module[module >= 0])
+    order = live_ids[np.argsort(sizes[live_ids], kind="stable")]
+    remap = np.full(n_modules, -1, np.int64)
+    remap[order] = np.arange(len(order))
+    module = np.where(module >= 0, remap[np.clip(module, 0, None)], -1)
+    if dead_frac > 0:
+        # the published real run covers only 3,773 of 7,523 genes with
+        # paths (reference README.md:32) — about half the real expression
+        # genes never co-express above threshold. Kill a matching
+        # fraction outright (pure noise in BOTH classes): their walks are
+        # singletons in both group graphs and drop as common paths,
+        # instead of becoming conflicting one-class singletons.
+        rng = np.random.default_rng(seed + 13)
+        dead = rng.random(n_common) < dead_frac
+        module = np.where(dead, -1, module)
     expr_genes = [genes[i] for i in chosen]
     expr = synth.synth_expression(expr_genes, raw["labels"], module, seed,
                                   shared_frac=shared_frac, off_frac=off_frac)
