@@ -160,3 +160,51 @@ def lp_find_lgroups(mat: np.ndarray, gene_list: Sequence[str],
     out[cl == poor_cl] = 1
     out[cl == largest] = 2
     return out
+
+
+# ---------------------------------------------------------------- step 6
+def lp_minmax(scores: np.ndarray, lo: float = 0.0,
+              hi: float = 1.0) -> np.ndarray:
+    mn, mx = scores.min(), scores.max()
+    return (hi - lo) / (mx - mn) * (scores - mn) + lo
+
+
+def lp_tstat(x: np.ndarray, y: np.ndarray) -> float:
+    """Pooled-variance two-sample t (ddof=1), 0 on zero denominators."""
+    import math
+    sx, sy = x.std(ddof=1), y.std(ddof=1)
+    nx, ny = len(x), len(y)
+    d1 = math.sqrt(((nx - 1.0) * sx * sx + (ny - 1.0) * sy * sy)
+                   / float(nx + ny - 2))
+    d2 = math.sqrt(1.0 / nx + 1.0 / ny)
+    if d1 > 0.0 and d2 > 0.0:
+        return (x.mean() - y.mean()) / d1 / d2
+    return 0.0
+
+
+def lp_select_biomarkers(mat: np.ndarray, lgroups: np.ndarray,
+                         expr: np.ndarray, labels: np.ndarray,
+                         gene_list: Sequence[str],
+                         num_biomarker: int) -> List[str]:
+    """Main-inline step 6: per L-group (0 good then 1 poor), gene score =
+    0.5*(minmax row-norm + minmax |pooled-t|); Python stable sort
+    DESCENDING by score, top-N genes sorted by symbol; the two lists
+    concatenate and the union sorts again."""
+    from operator import itemgetter
+
+    gene_arr = np.asarray(gene_list)
+    out: List[str] = []
+    for grp in (0, 1):
+        sub_mat = mat[lgroups == grp]
+        sub_genes = gene_arr[lgroups == grp]
+        sub_expr = expr[:, lgroups == grp]
+        d = lp_minmax(np.linalg.norm(sub_mat, axis=1), 0.0, 1.0)
+        t = np.zeros(sub_expr.shape[1], dtype=np.float32)
+        for i in range(sub_expr.shape[1]):
+            t[i] = abs(lp_tstat(sub_expr[labels == 0, i],
+                                sub_expr[labels == 1, i]))
+        score = 0.5 * (d + lp_minmax(t, 0.0, 1.0))
+        pairs = sorted(zip(sub_genes.tolist(), score.tolist()),
+                       key=itemgetter(1), reverse=True)
+        out += sorted(g for g, _s in pairs[:num_biomarker])
+    return sorted(out)

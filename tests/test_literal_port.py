@@ -184,3 +184,22 @@ def test_walk_distribution_matches_port(graphs):
     # coverage: every gene visited comparably often
     r = np.corrcoef(port_cov, fw_cov)[0, 1]
     assert r > 0.98, f"coverage correlation {r:.4f}"
+
+
+def test_biomarker_selection_matches_port():
+    """Step 6 (d-score + t-score + top-N union) — framework
+    select_biomarkers vs the literal port, exact list equality."""
+    from g2vec_amd.scoring import select_biomarkers
+    from literal_port import lp_select_biomarkers
+
+    rng = np.random.default_rng(17)
+    G, S, h = 300, 60, 16
+    W = rng.standard_normal((G, h)).astype(np.float32)
+    expr = rng.standard_normal((S, G)).astype(np.float32)
+    labels = (rng.random(S) < 0.45).astype(np.int64)
+    lg = rng.integers(0, 3, size=G)
+    genes = [f"SYM{i:04d}" for i in range(G)]
+    for n in (10, 50, 500):
+        want = lp_select_biomarkers(W, lg, expr, labels, genes, n)
+        got = select_biomarkers(W, lg, expr, labels, genes, n)
+        assert got == want, n
