@@ -175,3 +175,37 @@ def test_kgranular_early_stop_matches_per_epoch_on_gpu():
     assert stop == res.stop_epoch
     assert hist == pytest.approx(res.acc_val_history, abs=0)
     assert torch.equal(W.cpu(), res.W_ih.cpu())
+
+
+@pytest.mark.timeout(600)
+def test_rccl_allreduce_capturable_in_hipgraph():
+    """The SCALE path captures the grad all-reduce inside epoch graphs
+    at world>1 (behind DistContext.graph_capture_ok). Multi-rank needs
+    multiple GPUs, but a WORLD=1 RCCL process group still launches real
+    RCCL kernels — capture one into a hipGraph, replay it, and verify:
+    this exercises the same RCCL stream-capture machinery the probe
+    relies on."""
+    import os
+    import socket
+
+    import torch.distributed as dist
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    dist.init_process_group("nccl", rank=0, world_size=1,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        t = torch.ones(64, device="cuda")
+        dist.all_reduce(t)              # connect communicator
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g, capture_error_mode="thread_local"):
+            dist.all_reduce(t)
+            t.mul_(2.0)
+        for _ in range(3):
+            g.replay()
+        torch.cuda.synchronize()
+        assert torch.allclose(t, torch.full_like(t, 8.0))
+    finally:
+        dist.destroy_process_group()
