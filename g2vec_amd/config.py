@@ -58,6 +58,17 @@ class G2VecConfig:
     load_paths: str = ""
     log_jsonl: str = ""             # structured metrics sink
     deterministic_grads: bool = False  # bitwise-reproducible dW_ih reduction (no atomics)
+    gene_relabel: str = "auto"      # trainer-internal gene-id relabeling for
+                                    # gather locality: "on" | "off" | "auto"
+                                    # (auto = on at n_genes >= 100k, where the
+                                    # random s-gathers hit the L2 random-line
+                                    # wall). First-touch order over the path
+                                    # set makes co-path genes contiguous, so
+                                    # consecutive paths read L1-resident
+                                    # s/W slices. Pure layout change: weights
+                                    # are un-permuted before results leave the
+                                    # trainer; outputs are unchanged up to
+                                    # fp32 reduction order.
     use_hipgraph: bool = True       # record the full-batch epoch into a hipGraph
 
     def validate(self) -> None:
@@ -89,6 +100,8 @@ class G2VecConfig:
             raise ValueError(f"bad trainer_path {self.trainer_path}")
         if self.kmeans_backend not in ("auto", "sklearn", "torch"):
             raise ValueError(f"bad kmeans_backend {self.kmeans_backend}")
+        if self.gene_relabel not in ("auto", "on", "off"):
+            raise ValueError(f"bad gene_relabel {self.gene_relabel}")
 
 
 def resolve_device(device: str) -> str:

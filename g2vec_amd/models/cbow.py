@@ -155,6 +155,32 @@ class CbowTrainer:
         order = torch.argsort(first, stable=True)
         return subset(ps, order)
 
+    def _relabel_enabled(self) -> bool:
+        mode = self.cfg.gene_relabel
+        return mode == "on" or (mode == "auto" and self.G >= 100_000)
+
+    def _first_touch_order(self, ps: PathSet) -> torch.Tensor:
+        """i64 [G] new-id -> old-id: genes ordered by their first
+        appearance in the flat path stream. Co-path genes (one walk ≈ one
+        co-expression module) land on contiguous new ids, so the fwd/eval
+        s-gathers and the W-row gathers of consecutive paths touch
+        L1-resident slices instead of random 4-byte lines across the
+        whole table — the measured wall of the 100k+-gene configs
+        (profiles/README.md, 1M-config notes). Untouched genes keep
+        their relative order at the end. Rank 0's order is broadcast so
+        every DP rank maps identically (the c all-reduce lives in the
+        relabeled id space)."""
+        big = torch.iinfo(torch.int64).max
+        first = torch.full((self.G,), big, dtype=torch.int64,
+                           device=ps.genes.device)
+        pos = torch.arange(ps.genes.numel(), dtype=torch.int64,
+                           device=ps.genes.device)
+        first.scatter_reduce_(0, ps.genes.long(), pos, reduce="amin")
+        # stable argsort: untouched genes (key=big) keep ascending old id
+        order = torch.argsort(first, stable=True)
+        self.ctx.broadcast_(order)
+        return order
+
     # ------------------------------------------------------------------ setup
     def setup(self, ps: PathSet, pre_sharded: bool = False):
         """Initialise weights/optimizer state and split the path set.
@@ -167,6 +193,20 @@ class CbowTrainer:
         W, who = self._init_weights(gen)
         self.ctx.broadcast_(W)      # C4: replicated params (all ranks identical)
         self.ctx.broadcast_(who)
+        self.gene_order = None      # new-id -> old-id (gather-locality relabel)
+        self.gene_o2n = None
+        if self._relabel_enabled():
+            order = self._first_touch_order(ps)
+            o2n = torch.empty_like(order)
+            o2n[order] = torch.arange(self.G, dtype=torch.int64,
+                                      device=order.device)
+            ps = PathSet(o2n[ps.genes.long()].int().contiguous(),
+                         ps.offsets, ps.labels, ps.n_genes)
+            # weights were drawn in OLD-id order (seed parity with the
+            # unrelabeled run): permute rows into the relabeled layout —
+            # gene old(j)=order[j] keeps its exact init vector
+            W = W[order].contiguous()
+            self.gene_order, self.gene_o2n = order, o2n
         tr, vl = self._split(ps, pre_sharded)
 
         use_general = cfg.trainer_path == "general"
@@ -707,7 +747,8 @@ class CbowTrainer:
             st.W_keep.copy_(st.W)   # keep-last-good snapshot (G2Vec.py:283)
         self.log("    Optimization Finish")
 
-        return TrainResult(W_ih=st.W_keep, stop_epoch=stop_epoch,
+        return TrainResult(W_ih=self._unrelabel(st.W_keep),
+                           stop_epoch=stop_epoch,
                            acc_val=acc_val, acc_tr=acc_tr,
                            epochs_run=epochs_run, acc_val_history=acc_hist,
                            epoch_times_s=epoch_times, wall_to_acc_s=wall_to_acc)
@@ -755,11 +796,19 @@ class CbowTrainer:
         else:
             acc_val, acc_tr = hist[-1], tr_hist[-1]
         self.log("    Optimization Finish")
-        return TrainResult(W_ih=W_final, stop_epoch=stop_epoch,
+        return TrainResult(W_ih=self._unrelabel(W_final),
+                           stop_epoch=stop_epoch,
                            acc_val=acc_val, acc_tr=acc_tr,
                            epochs_run=epochs_run, acc_val_history=hist,
                            epoch_times_s=epoch_times,
                            wall_to_acc_s=wall_box[0])
+
+    def _unrelabel(self, W: torch.Tensor) -> torch.Tensor:
+        """Rows back to original gene order (no-op when relabeling is off).
+        Applied once, where weights leave the trainer."""
+        if self.gene_o2n is None:
+            return W
+        return W[self.gene_o2n].contiguous()
 
     # ------------------------------------------------------------------ steps
     def _slice(self, ps: PathSet, lo: int, hi: int):

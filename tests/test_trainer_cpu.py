@@ -208,3 +208,36 @@ def test_kgranular_no_dip_runs_all_epochs():
     if stop == -1:
         assert hist == pytest.approx(res.acc_val_history, abs=0)
         assert torch.equal(W, res.W_ih)
+
+
+def test_gene_relabel_is_pure_layout():
+    """gene_relabel="on" (the gather-locality relabeling used at 100k+
+    genes) must be invisible in results: weights are drawn in original
+    gene order and permuted into the relabeled layout, so every gene
+    keeps its exact init vector and the returned W_ih / trajectory match
+    the unrelabeled run up to fp32 reduction order."""
+    ps = _random_pathset(G=80, P=300, seed=5)
+    results = {}
+    for mode in ("off", "on"):
+        cfg = G2VecConfig(hidden=64, epochs=10, early_stop=False, seed=3,
+                          device="cpu", dtype="fp32", gene_relabel=mode)
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                         log=lambda *a, **k: None)
+        results[mode] = tr.train(ps)
+    a, b = results["off"], results["on"]
+    assert b.acc_val_history == pytest.approx(a.acc_val_history, abs=1e-6)
+    assert torch.allclose(b.W_ih, a.W_ih, atol=1e-5)
+    assert b.W_ih.shape == a.W_ih.shape
+
+
+def test_gene_relabel_early_stop_unpermutes_kept_weights():
+    ps = _random_pathset(G=70, P=260, seed=11)
+    outs = {}
+    for mode in ("off", "on"):
+        cfg = G2VecConfig(hidden=64, epochs=40, early_stop=True, seed=2,
+                          device="cpu", dtype="fp32", gene_relabel=mode)
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                         log=lambda *a, **k: None)
+        outs[mode] = tr.train(ps)
+    assert outs["on"].stop_epoch == outs["off"].stop_epoch
+    assert torch.allclose(outs["on"].W_ih, outs["off"].W_ih, atol=1e-5)
