@@ -56,15 +56,26 @@ def cbow_fwd_scalar(s, genes, offsets, labels, inv_b: float, want_grad: bool):
 
 def cbow_eval_counts_(s, genes, offsets, labels, p_split: int,
                       counts: torch.Tensor, dO: Optional[torch.Tensor] = None,
-                      inv_b: float = 1.0) -> None:
+                      inv_b: float = 1.0, scan=None) -> None:
     """Accumulate the concatenated train+val correct counts into counts[2]
     (caller zeroes it). One fused kernel on GPU; oracle math on CPU.
 
     dO given: also emit the train split's (p < p_split) dlogit into dO with
     scale inv_b — the fused next-epoch forward (this eval's s is the next
-    epoch's pre-update s), bitwise-identical to a separate cbow_fwd_scalar
-    over the train paths."""
+    epoch's pre-update s), numerically identical to a separate
+    cbow_fwd_scalar over the train paths (fp32 summation order differs
+    between the kernel variants).
+
+    scan = (pathid i32[nnz], piece f32[>=P*cap], cap): persistent buffers
+    enabling the instance-parallel segmented-scan variant — the hot
+    steady-state eval (see eval_scan_kernel)."""
     if s.is_cuda:
+        if scan is not None:
+            pathid, piece, cap = scan
+            native().cbow_eval_scan_(s, genes, pathid, offsets, labels,
+                                     int(p_split), int(cap), piece, counts,
+                                     dO=dO, inv_b=float(inv_b))
+            return
         native().cbow_eval_counts_(s, genes, offsets, labels, int(p_split),
                                    counts, dO=dO, inv_b=float(inv_b))
         return

@@ -133,6 +133,49 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   LAUNCH_CHECK();
 }
 
+void cbow_eval_scan_(torch::Tensor s, torch::Tensor genes,
+                     torch::Tensor pathid, torch::Tensor offs,
+                     torch::Tensor labels, int64_t p_split, int64_t cap,
+                     torch::Tensor piece, torch::Tensor counts,
+                     std::optional<torch::Tensor> dO, double inv_b) {
+  // instance-parallel fused eval (see eval_scan_kernel): piece buffer is
+  // persistent (hipGraph-stable) and sized P * cap
+  CHECK_DEV(s); CHECK_CONT(s); CHECK_F32(s);
+  CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
+  CHECK_DEV(pathid); CHECK_CONT(pathid); CHECK_I32(pathid);
+  CHECK_DEV(offs); CHECK_CONT(offs); CHECK_I32(offs);
+  CHECK_DEV(labels); CHECK_CONT(labels); CHECK_F32(labels);
+  CHECK_DEV(piece); CHECK_CONT(piece); CHECK_F32(piece);
+  CHECK_DEV(counts); CHECK_CONT(counts); CHECK_F32(counts);
+  const long long P = labels.numel();
+  const long long nnz = genes.numel();
+  TORCH_CHECK(piece.numel() >= P * cap, "piece buffer too small");
+  if (dO) {
+    CHECK_DEV(*dO); CHECK_CONT(*dO); CHECK_F32(*dO);
+    TORCH_CHECK(dO->numel() >= p_split, "dO must cover the train split");
+  }
+  if (P == 0) return;
+  const int grid_a = grid_for(nnz, 256);
+  hipLaunchKernelGGL(eval_scan_kernel, dim3(grid_a), dim3(256), 0,
+                     cur_stream(), s.data_ptr<float>(),
+                     genes.data_ptr<int>(), pathid.data_ptr<int>(),
+                     offs.data_ptr<int>(), nnz, (int)cap,
+                     piece.data_ptr<float>());
+  int grid_b = grid_for(P, 256);
+  if (grid_b > 2048) grid_b = 2048;
+  auto partials = torch::empty({grid_b, 2},
+      torch::TensorOptions().dtype(at::kFloat).device(s.device()));
+  hipLaunchKernelGGL(eval_finish_kernel, dim3(grid_b), dim3(256), 0,
+                     cur_stream(), piece.data_ptr<float>(),
+                     offs.data_ptr<int>(), labels.data_ptr<float>(), P,
+                     (long long)p_split, (int)cap, partials.data_ptr<float>(),
+                     dO ? dO->data_ptr<float>() : nullptr, (float)inv_b);
+  hipLaunchKernelGGL(fold_partials_kernel, dim3(1), dim3(256), 0,
+                     cur_stream(), partials.data_ptr<float>(), grid_b,
+                     counts.data_ptr<float>());
+  LAUNCH_CHECK();
+}
+
 void scatter_dO_det_(torch::Tensor inst_path, torch::Tensor seg_start,
                      torch::Tensor seg_gene, torch::Tensor dO,
                      torch::Tensor c) {
@@ -497,6 +540,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO");
   m.def("scatter_dO_det_", &scatter_dO_det_,
         "deterministic c += X^T dO (one slab, out-arg)");
+  m.def("cbow_eval_scan_", &cbow_eval_scan_,
+        "instance-parallel fused eval (segmented scan + finish)");
   m.def("cbow_eval_counts_", &cbow_eval_counts_,
         py::arg("s"), py::arg("genes"), py::arg("offs"), py::arg("labels"),
         py::arg("p_split"), py::arg("counts"),

@@ -871,3 +871,85 @@ extern "C" __global__ void trunc_normal_kernel(float* __restrict__ out,
     out[i] = x * std;
   }
 }
+
+// ================================= fast path: instance-parallel fused eval
+// The subwave-per-path eval kernel is latency-bound: ~21-gene paths give
+// each 16-lane group a 2-iteration dependent gather chain and idle lanes
+// (26 us over 1.5M instances at ex_* scale). This pair streams the flat
+// CSR instance space instead: every lane gathers ONE s value, a 64-lane
+// segmented inclusive scan (head flags at path boundaries) produces
+// per-path piece sums per 64-instance window, and a thread-per-path
+// finish kernel folds each path's <= cap pieces (fixed ascending window
+// order -> deterministic), computes correctness counts for the two
+// splits, and emits the train split's next-epoch dlogits (same fusion
+// as cbow_eval_counts_kernel). Wave windows are aligned: every wave
+// covers instances [w*64, w*64+64).
+extern "C" __global__ void __launch_bounds__(256)
+eval_scan_kernel(const float* __restrict__ s, const int* __restrict__ genes,
+                 const int* __restrict__ pathid, const int* __restrict__ offs,
+                 long long nnz, int cap, float* __restrict__ piece) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i - lane < nnz; i += stride) {           // whole wave iterates together
+    const bool act = i < nnz;
+    const int pid = act ? pathid[i] : -1;
+    float val = act ? s[genes[i]] : 0.f;
+    const int pid_up = __shfl_up(pid, 1);
+    int f = (lane == 0) || (pid != pid_up);     // head of a window piece
+    const int head = f;
+#pragma unroll
+    for (int o = 1; o < WAVE; o <<= 1) {        // segmented inclusive scan
+      const float tv = __shfl_up(val, o);
+      const int tf = __shfl_up(f, o);
+      if (lane >= o) {
+        if (!f) val += tv;
+        f = f | tf;
+      }
+    }
+    const int head_down = __shfl_down(head, 1);
+    const bool last = (lane == WAVE - 1) || head_down;
+    if (act && last) {
+      const long long w = i >> 6;
+      const long long w0 = (long long)offs[pid] >> 6;
+      piece[(long long)pid * cap + (w - w0)] = val;
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+eval_finish_kernel(const float* __restrict__ piece, const int* __restrict__ offs,
+                   const float* __restrict__ labels, long long P,
+                   long long p_split, int cap, float* __restrict__ partials,
+                   float* __restrict__ dO, float inv_b) {
+  float c0 = 0.f, c1 = 0.f;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       p < P; p += stride) {
+    const int lo = offs[p], hi = offs[p + 1];
+    const long long w0 = (long long)lo >> 6;
+    const long long w1 = ((long long)hi - 1) >> 6;
+    float o = 0.f;
+    const float* pp = piece + (long long)p * cap;
+    for (long long w = w0; w <= w1; ++w) o += pp[w - w0];
+    const float y = labels[p];
+    const float corr = (((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f);
+    if (p < p_split) {
+      c0 += corr;
+      if (dO) dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
+    } else {
+      c1 += corr;
+    }
+  }
+  c0 = wave_sum(c0);
+  c1 = wave_sum(c1);
+  __shared__ float sm[8];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  if (lane == 0) { sm[wib] = c0; sm[4 + wib] = c1; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    partials[2 * blockIdx.x] = sm[0] + sm[1] + sm[2] + sm[3];
+    partials[2 * blockIdx.x + 1] = sm[4] + sm[5] + sm[6] + sm[7];
+  }
+}

@@ -277,3 +277,56 @@ def test_walks_bitwise_high_degree_paths():
     assert torch.equal(gl.cpu(), cl)
     assert torch.equal(gn.cpu(), cn)
     assert torch.equal(gh.cpu(), ch)
+
+
+@pytest.mark.timeout(600)
+def test_eval_scan_matches_subwave_kernel_and_oracle():
+    """Instance-parallel segmented-scan eval (eval_scan/finish kernels)
+    vs the subwave-per-path kernel and the CPU oracle: identical correct
+    counts, dO within fp32 reduction-order tolerance. Covers paths that
+    span multiple 64-instance windows and window-straddling boundaries."""
+    import numpy as np
+
+    from g2vec_amd.ops import cpu_ref
+    rng = np.random.default_rng(31)
+    G, P = 500, 3000
+    genes, offs, labels = [], [0], []
+    for p in range(P):
+        # mix of tiny and window-spanning paths (up to 150 genes)
+        L = int(rng.integers(1, 150)) if p % 7 == 0 else int(rng.integers(1, 25))
+        L = min(L, G)
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda")
+    s = torch.randn(G, device=dev) * 0.3
+    g_t = torch.tensor(genes, dtype=torch.int32, device=dev)
+    o_t = torch.tensor(offs, dtype=torch.int32, device=dev)
+    l_t = torch.tensor(labels, dtype=torch.float32, device=dev)
+    p_split = P * 4 // 5
+    inv_b = 1.0 / p_split
+
+    counts_a = torch.zeros(2, device=dev)
+    dO_a = torch.zeros(p_split, device=dev)
+    ops.native().cbow_eval_counts_(s, g_t, o_t, l_t, p_split, counts_a,
+                                   dO=dO_a, inv_b=inv_b)
+
+    lens = (o_t[1:] - o_t[:-1]).long()
+    pathid = torch.repeat_interleave(
+        torch.arange(P, dtype=torch.int32, device=dev), lens)
+    cap = 150 // 64 + 2
+    piece = torch.empty(P * cap, dtype=torch.float32, device=dev)
+    counts_b = torch.zeros(2, device=dev)
+    dO_b = torch.zeros(p_split, device=dev)
+    ops.native().cbow_eval_scan_(s, g_t, pathid, o_t, l_t, p_split, cap,
+                                 piece, counts_b, dO=dO_b, inv_b=inv_b)
+
+    assert torch.equal(counts_a.cpu(), counts_b.cpu())
+    assert torch.allclose(dO_a, dO_b, atol=1e-6)
+
+    # oracle cross-check
+    _l, corr, d = cpu_ref.cbow_fwd_scalar(s.cpu(), g_t.cpu(), o_t.cpu(),
+                                          l_t.cpu(), inv_b, True)
+    assert float(counts_b[0]) == float(corr[:p_split].sum())
+    assert float(counts_b[1]) == float(corr[p_split:].sum())
+    assert torch.allclose(dO_b.cpu(), d[:p_split], atol=1e-5)
