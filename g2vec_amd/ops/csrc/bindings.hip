@@ -541,6 +541,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_dO_det_", &scatter_dO_det_,
         "deterministic c += X^T dO (one slab, out-arg)");
   m.def("cbow_eval_scan_", &cbow_eval_scan_,
+        py::arg("s"), py::arg("genes"), py::arg("pathid"), py::arg("offs"),
+        py::arg("labels"), py::arg("p_split"), py::arg("cap"),
+        py::arg("piece"), py::arg("counts"),
+        py::arg("dO") = py::none(), py::arg("inv_b") = 1.0,
         "instance-parallel fused eval (segmented scan + finish)");
   m.def("cbow_eval_counts_", &cbow_eval_counts_,
         py::arg("s"), py::arg("genes"), py::arg("offs"), py::arg("labels"),
