@@ -155,12 +155,21 @@ void cbow_eval_scan_(torch::Tensor s, torch::Tensor genes,
     TORCH_CHECK(dO->numel() >= p_split, "dO must cover the train split");
   }
   if (P == 0) return;
-  const int grid_a = grid_for(nnz, 256);
-  hipLaunchKernelGGL(eval_scan_kernel, dim3(grid_a), dim3(256), 0,
+  int grid_a = grid_for(nnz, 256);
+  const long long G = s.numel();
+  int g_lds = 0;
+  size_t lds = 0;
+  if (G * 4 <= 48 * 1024) {
+    // LDS-staged s: persistent blocks amortize the one-time 30 KB stage
+    g_lds = (int)G;
+    lds = (size_t)G * sizeof(float);
+    if (grid_a > 1536) grid_a = 1536;
+  }
+  hipLaunchKernelGGL(eval_scan_kernel, dim3(grid_a), dim3(256), lds,
                      cur_stream(), s.data_ptr<float>(),
                      genes.data_ptr<int>(), pathid.data_ptr<int>(),
                      offs.data_ptr<int>(), nnz, (int)cap,
-                     piece.data_ptr<float>());
+                     piece.data_ptr<float>(), g_lds);
   int grid_b = grid_for(P, 256);
   if (grid_b > 2048) grid_b = 2048;
   auto partials = torch::empty({grid_b, 2},

@@ -887,14 +887,26 @@ extern "C" __global__ void trunc_normal_kernel(float* __restrict__ out,
 extern "C" __global__ void __launch_bounds__(256)
 eval_scan_kernel(const float* __restrict__ s, const int* __restrict__ genes,
                  const int* __restrict__ pathid, const int* __restrict__ offs,
-                 long long nnz, int cap, float* __restrict__ piece) {
+                 long long nnz, int cap, float* __restrict__ piece,
+                 int g_lds) {
+  // g_lds > 0: the whole s table is staged into LDS once per
+  // (persistent, grid-strided) block — random 4-byte s-gathers become
+  // ds_read ops instead of 64-way-divergent L1 line lookups, the
+  // measured wall of the gather at <= 48 KB tables
+  extern __shared__ float s_lds[];
+  const float* st = s;
+  if (g_lds > 0) {
+    for (int g = threadIdx.x; g < g_lds; g += blockDim.x) s_lds[g] = s[g];
+    __syncthreads();
+    st = s_lds;
+  }
   const int lane = threadIdx.x & (WAVE - 1);
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i - lane < nnz; i += stride) {           // whole wave iterates together
     const bool act = i < nnz;
     const int pid = act ? pathid[i] : -1;
-    float val = act ? s[genes[i]] : 0.f;
+    float val = act ? st[genes[i]] : 0.f;
     const int pid_up = __shfl_up(pid, 1);
     int f = (lane == 0) || (pid != pid_up);     // head of a window piece
     const int head = f;
