@@ -17,7 +17,8 @@ def main():
     rng = np.random.default_rng(0)
     G = int(sys.argv[1]) if len(sys.argv) > 1 else 7523
     P = int(sys.argv[2]) if len(sys.argv) > 2 else 75000
-    lens = rng.integers(2, 40, size=P)
+    lo = int(os.environ.get('LMIN', 2)); hi = int(os.environ.get('LMAX', 40))
+    lens = rng.integers(lo, hi, size=P)
     genes = np.concatenate([rng.integers(0, G, size=n) for n in lens])
     offs = np.concatenate([[0], np.cumsum(lens)])
     labels = rng.integers(0, 2, size=P).astype(np.float32)
@@ -30,7 +31,7 @@ def main():
     lens_t = (o_t[1:] - o_t[:-1]).long()
     pathid = torch.repeat_interleave(
         torch.arange(P, dtype=torch.int32, device=dev), lens_t)
-    cap = 40 // 64 + 2
+    cap = (hi - 1) // 64 + 2
     piece = torch.empty(P * cap, dtype=torch.float32, device=dev)
     counts = torch.zeros(2, device=dev)
     dO = torch.zeros(p_split, device=dev)
