@@ -253,22 +253,15 @@ class CbowTrainer:
             st.ev_genes = torch.cat([tr.genes, vl.genes]).contiguous()
             st.ev_offsets = torch.cat([tr.offsets, off_vl[1:]]).contiguous()
             st.ev_labels = torch.cat([tr.labels, vl.labels]).contiguous()
-            st.ev_scan = None
-            if self.device.type == "cuda":
-                # instance-parallel eval buffers: per-instance path id +
-                # the per-window piece table (persistent: hipGraph-stable)
-                P_ev = int(st.ev_labels.numel())
-                lens = (st.ev_offsets[1:] - st.ev_offsets[:-1]).long()
-                pathid = torch.repeat_interleave(
-                    torch.arange(P_ev, dtype=torch.int32,
-                                 device=self.device), lens)
-                cap = cfg.len_path // 64 + 2
-                piece = torch.empty(P_ev * cap, dtype=torch.float32,
-                                    device=self.device)
-                st.ev_scan = (pathid.contiguous(), piece, cap)
+            # NOTE: an instance-parallel segmented-scan eval
+            # (eval_scan_kernel) was built and A/B-measured against the
+            # subwave-per-path kernel — slower at every tested shape
+            # (44 vs 21.5 us at the ex_* path-length distribution, tie at
+            # 1M genes; tools/bench_eval.py, profiles/README.md). The
+            # subwave kernel stays the epoch-body eval; the scan pair
+            # remains available + tested as the research alternative.
         else:
             st.ev_genes = None
-            st.ev_scan = None
         # full-batch steady state fuses the forward into the PREVIOUS
         # epoch's eval (same s, same train paths — the gather runs once per
         # weight version): dO_buf carries the pending dlogits; seed it here
@@ -328,8 +321,7 @@ class CbowTrainer:
         if st.ev_genes is not None:
             ops.cbow_eval_counts_(st.s_buf, st.ev_genes, st.ev_offsets,
                                   st.ev_labels, tr.n_paths, counts_out,
-                                  dO=st.dO_buf, inv_b=st.inv_b,
-                                  scan=st.ev_scan)
+                                  dO=st.dO_buf, inv_b=st.inv_b)
         if reduce_counts:
             self.ctx.allreduce_(counts_out)     # C3: 2-float metric reduce
 
