@@ -185,7 +185,6 @@ def test_rccl_allreduce_capturable_in_hipgraph():
     RCCL kernels — capture one into a hipGraph, replay it, and verify:
     this exercises the same RCCL stream-capture machinery the probe
     relies on."""
-    import os
     import socket
 
     import torch.distributed as dist

@@ -21,7 +21,7 @@ from g2vec_amd.utils import synth
 from g2vec_amd.walks import WalkSet
 
 from literal_port import (lp_adj_matrix, lp_find_lgroups, lp_gene_freq,
-                          lp_integrate, lp_pathset, lp_random_path)
+                          lp_integrate, lp_random_path)
 
 
 @pytest.fixture(scope="module")

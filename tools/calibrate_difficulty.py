@@ -17,7 +17,6 @@ import argparse
 import sys
 import time
 
-import numpy as np
 import torch
 
 import os

@@ -1,6 +1,5 @@
 #!/usr/bin/env python3
 """Micro-diagnosis of the slow integrate ops (bincount, masked index)."""
-import sys
 import time
 
 import torch
