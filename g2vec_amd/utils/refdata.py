@@ -103,15 +103,14 @@ def make_real_dataset(seed: int = 0, n_common: int = N_COMMON_PUBLISHED,
                       ref_dir: Optional[str] = None) -> Dict:
     """Real network + real clinical + synthesized expression.
 
-    Difficulty note: with the default knobs (calibrated sweep in
-    tools/calibrate_difficulty.py's real-topology variant) the seeded
-    val-ACC climbs ~0.58 -> ~0.83 and plateaus BELOW the published
-    0.8837 — every gene is walked from in both group graphs, and on the
-    real hub-dense topology the off-class walks over a module's genes
-    are genuinely label-ambiguous for the linear model. The real-data
-    variant's job is the real TOPOLOGY (hub rows >256, real loaders,
-    count invariants); the published-difficulty convergence headline
-    lives on the calibrated synthetic ex_* config (bench.py defaults).
+    Difficulty note: with the default knobs (size-ordered community ids,
+    shared=0.1, off=0.45 — calibrated in the real-topology sweep) the
+    seeded val-ACC climbs ~0.58 -> ~0.88 at the full 10-repetition walk
+    budget: the bench --real-data probe measured best-ACC 0.8825 with
+    3/3 training seeds crossing 0.88 (profiles/bench_real_r2.json). At
+    reduced walk budgets (reps <= 3) the ceiling sits lower (~0.83):
+    off-class walks over a module's genes are label-ambiguous for the
+    linear model until enough paths accumulate.
 
     Returns {'expr' f32 [S, n_common] (samples x chosen genes),
     'expr_genes', 'samples', 'labels', 'net_genes', 'edge_idx' (full real
