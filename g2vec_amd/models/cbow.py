@@ -715,7 +715,11 @@ class CbowTrainer:
         st = self.setup(ps, pre_sharded)
 
         if (cfg.trainer_path != "general" and cfg.batch_size == 0 and
-                self.device.type == "cuda"):
+                (self.device.type == "cuda" or
+                 (cfg.early_stop and cfg.earlystop_every > 1))):
+            # GPU fast path always; CPU too when --earlystop-every opts
+            # into the k-granular runner (the flag must never be a
+            # silent no-op — cf. the round-1 --dtype finding)
             return self._train_pipelined(st)
 
         before_val, before_tr = -1.0, -1.0

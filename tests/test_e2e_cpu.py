@@ -185,3 +185,34 @@ def test_load_model_resumes_without_training(tiny_files, tmp_path):
     bad.hidden = 256
     with pytest.raises(ValueError, match="load-model"):
         run(bad)
+
+
+def test_earlystop_every_cli_plumb(tiny_files, tmp_path):
+    """--earlystop-every K must plumb through the CLI and reproduce the
+    per-epoch early-stop results exactly (k-granular replay semantics),
+    never acting as a silent no-op."""
+    import subprocess
+    import sys as _sys
+
+    outs = {}
+    for tag, extra in (("k1", []), ("k4", ["--earlystop-every", "4"])):
+        r = subprocess.run(
+            [_sys.executable, "-m", "g2vec_amd",
+             tiny_files["expression"], tiny_files["clinical"],
+             tiny_files["network"], str(tmp_path / tag),
+             "-p", "12", "-r", "2", "-e", "30", "--seed", "0",
+             "--device", "cpu"] + extra,
+            capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stderr[-1500:]
+        outs[tag] = r.stdout
+        assert (tmp_path / f"{tag}_vectors.txt").exists()
+    # identical epoch transcript lines (trajectory + stop epoch)
+    lines1 = [l for l in outs["k1"].splitlines() if "Epoch" in l]
+    lines4 = [l for l in outs["k4"].splitlines() if "Epoch" in l]
+    def strip_t(ls):
+        import re
+        return [re.sub(r"\(\d+\.\d+ sec\)", "", l) for l in ls]
+    assert strip_t(lines1) == strip_t(lines4)
+    v1 = (tmp_path / "k1_vectors.txt").read_text()
+    v4 = (tmp_path / "k4_vectors.txt").read_text()
+    assert v1 == v4
