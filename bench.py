@@ -87,6 +87,10 @@ def main() -> int:
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--trainer-path", choices=["fast", "general"],
                     default="fast")
+    ap.add_argument("--dtype", choices=["fp32", "bf16", "fp16"],
+                    default="fp32",
+                    help="W_ih gather storage dtype (general path only; "
+                         "the fast path computes fp32 regardless)")
     ap.add_argument("--n-genes", type=int, default=7523,
                     help="scale configs: 50000 / 200000 / 1000000")
     ap.add_argument("--n-edges", type=int, default=298799)
@@ -127,7 +131,7 @@ def main() -> int:
     cfg = G2VecConfig(hidden=args.hidden, len_path=args.len_path,
                       num_repetition=args.reps, epochs=500, seed=args.seed,
                       device=str(device.type), trainer_path=args.trainer_path,
-                      use_hipgraph=not args.no_hipgraph)
+                      dtype=args.dtype, use_hipgraph=not args.no_hipgraph)
 
     # ---- dataset + graphs + walks (input pipeline; measured, not the metric)
     if args.real_data:
@@ -267,7 +271,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_PATHS_PER_SEC, 2),
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": (f"real ex_NETWORK topology ({n_genes} common genes of "
                      f"9904/135 real clinical samples; synthetic expression; "
                      f"random-init weights)" if args.real_data else
