@@ -229,10 +229,27 @@ class CbowTrainer:
         # gene-sorted instance plan: drives the deterministic backward on
         # both trainer paths (genes never change across epochs)
         st.plan = ops.build_scatter_plan(tr.genes, tr.offsets, self.G)
+        if self.n_tr_global + self.n_vl_global == 0:
+            raise ValueError(
+                "no paths to train on: the integrated path set is empty "
+                "(every walk was dropped as a cross-group duplicate, or "
+                "the PCC-thresholded graphs have no edges — check the "
+                "expression signal / --pcc-threshold)")
         st.inv_b = 1.0 / max(self.n_tr_global, 1)
         P_loc = tr.n_paths
-        bs = cfg.batch_size if cfg.batch_size > 0 else P_loc
-        st.batches = [(i, min(i + bs, P_loc)) for i in range(0, max(P_loc, 1), bs)]
+        bs = cfg.batch_size if cfg.batch_size > 0 else max(P_loc, 1)
+        st.batches = [(i, min(i + bs, P_loc)) for i in range(0, P_loc, bs)]
+        if self.ctx.world > 1 and cfg.batch_size > 0:
+            # lockstep minibatching: strided shards can differ by one
+            # path, so per-rank batch COUNTS can differ — pad with empty
+            # batches up to the global max so every rank joins the same
+            # number of grad all-reduces (an empty batch contributes a
+            # zero gradient)
+            nb = torch.tensor([float(len(st.batches))], dtype=torch.float64,
+                              device=self.device)
+            self.ctx.allreduce_max_(nb)
+            while len(st.batches) < int(nb.item()):
+                st.batches.append((P_loc, P_loc))
         st.t_adam = 0
         st.epoch_idx = 0
         # persistent fast-path buffers (stable addresses across hipGraph replays)
@@ -354,8 +371,10 @@ class CbowTrainer:
             return acc_tr, acc_val
 
         for (lo, hi) in st.batches:
-            b_inv = st.inv_b if cfg.batch_size == 0 else 1.0 / (
-                (hi - lo) * self.ctx.world)
+            # empty pad batch (lockstep minibatching): zero local grad,
+            # scale irrelevant
+            b_inv = (st.inv_b if cfg.batch_size == 0 else
+                     (1.0 / ((hi - lo) * self.ctx.world) if hi > lo else 0.0))
             st.t_adam += 1
             if cfg.trainer_path == "general":
                 self._step_general(st, lo, hi, b_inv, st.t_adam)

@@ -88,6 +88,13 @@ def run(cfg: G2VecConfig, ctx: Optional[DistContext] = None) -> Dict:
         edge_idx = pp.edges_to_indices(network["edge"], common)
     n_samples, n_genes = data["expr"].shape
     n_edges = len(network["edge"])
+    n_poor = int(np.sum(np.asarray(data["label"]) == 1))
+    if min(n_poor, n_samples - n_poor) < 2:
+        # per-group PCC and the pooled t-statistic need >= 2 samples per
+        # prognosis group; the reference nan-propagates here — fail loud
+        raise ValueError(
+            f"each prognosis group needs >= 2 samples: clinical file has "
+            f"{n_samples - n_poor} good / {n_poor} poor")
     log("    n_samples: %d" % n_samples)
     log("    n_genes  : %d\t(common genes in both EXPRESSION and NETWORK)" % n_genes)
     log("    n_edges  : %d\t(edges with the common genes)" % n_edges)

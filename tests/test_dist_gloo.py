@@ -295,3 +295,40 @@ def test_dp2_kgranular_early_stop_parity_and_schedule():
     # reduce per block (+ replay grads on stop) — strictly fewer than
     # the 2-per-epoch per-epoch schedule
     assert n_ar < 2 * n_hist
+
+
+def _minibatch_uneven_worker(rank, world, port, out):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        # P=161: strided shards of the 128-path train split differ by one
+        # path, so per-rank batch counts differ -> the lockstep padding
+        # must keep the collective schedule aligned (no hang)
+        ps = _pathset(G=40, P=161, seed=13)
+        cfg = G2VecConfig(hidden=64, epochs=4, early_stop=False, seed=4,
+                          device="cpu", dtype="fp32", batch_size=33)
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                         log=lambda *a, **k: None)
+        res = tr.train(ps)
+        assert np.isfinite(res.acc_val)
+        if rank == 0:
+            out.put(res.acc_val)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_minibatch_uneven_shards_no_deadlock():
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    out = ctxm.Queue()
+    procs = [ctxm.Process(target=_minibatch_uneven_worker,
+                          args=(r, 2, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    acc = out.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert 0.0 <= acc <= 1.0

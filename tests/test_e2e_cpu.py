@@ -216,3 +216,31 @@ def test_earlystop_every_cli_plumb(tiny_files, tmp_path):
     v1 = (tmp_path / "k1_vectors.txt").read_text()
     v4 = (tmp_path / "k4_vectors.txt").read_text()
     assert v1 == v4
+
+
+def test_degenerate_inputs_fail_loudly(tiny_files, tmp_path):
+    """Empty prognosis group and empty path set raise typed errors with
+    actionable messages instead of nan-propagating (the reference's
+    behavior) or crashing obscurely."""
+    # (a) one-class clinical file
+    import shutil
+    bad_cli = tmp_path / "bad_CLINICAL.txt"
+    lines = open(tiny_files["clinical"]).read().splitlines()
+    with open(bad_cli, "w") as f:
+        f.write(lines[0] + "\n")
+        for ln in lines[1:]:
+            f.write(ln.split("\t")[0] + "\t0\n")
+    cfg = _cfg(tiny_files, tmp_path, clinical_file=str(bad_cli))
+    with pytest.raises(ValueError, match="prognosis group"):
+        run(cfg)
+
+    # (b) empty path set reaching the trainer
+    from g2vec_amd.models.cbow import CbowTrainer
+    from g2vec_amd.paths import PathSet
+    ps = PathSet(torch.zeros(0, dtype=torch.int32),
+                 torch.zeros(1, dtype=torch.int32),
+                 torch.zeros(0), 10)
+    tr = CbowTrainer(G2VecConfig(hidden=64, device="cpu"), 10,
+                     torch.device("cpu"), log=lambda *a, **k: None)
+    with pytest.raises(ValueError, match="no paths"):
+        tr.train(ps)
