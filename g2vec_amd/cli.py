@@ -51,6 +51,13 @@ def build_parser() -> argparse.ArgumentParser:
                    help="reproduce the shipped L-group disambiguation bug "
                         "(G2Vec.py:186-194, SURVEY 2.9)")
     p.add_argument("--no-early-stop", action="store_true")
+    p.add_argument("--train-ckpt", default="",
+                   help="mid-training checkpoint file (weights + Adam "
+                        "moments + early-stop trackers), written every "
+                        "--train-ckpt-every epochs; resume with "
+                        "--resume-train for an exact continuation")
+    p.add_argument("--train-ckpt-every", type=int, default=5)
+    p.add_argument("--resume-train", default="")
     p.add_argument("--earlystop-every", type=int, default=1,
                    help="read early-stop accuracies every K epochs and "
                         "deterministically replay to the dip on stop "
@@ -81,6 +88,8 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         kmeans_backend=a.kmeans, trainer_path=a.trainer_path,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, earlystop_every=a.earlystop_every,
+        train_ckpt=a.train_ckpt, train_ckpt_every=a.train_ckpt_every,
+        resume_train=a.resume_train,
         save_paths=a.save_paths,
         load_paths=a.load_paths, save_model=a.save_model,
         load_model=a.load_model,

@@ -54,6 +54,15 @@ class G2VecConfig:
                                     # "general": full gather/scatter kernel chain (K1-K8)
     save_paths: str = ""            # cache generated path set (de-facto checkpoint)
     save_model: str = ""            # save trained W_ih/W_ho + metadata (.pt)
+    train_ckpt: str = ""            # mid-TRAINING checkpoint file: weights +
+                                    # Adam moments + early-stop trackers,
+                                    # written every train_ckpt_every epochs;
+                                    # --resume-train continues the EXACT
+                                    # trajectory (epoch bodies deterministic).
+                                    # Runs the synchronous epoch loop
+                                    # (checkpoints are epoch-aligned).
+    train_ckpt_every: int = 5
+    resume_train: str = ""          # resume training from a train_ckpt file
     load_model: str = ""            # resume: skip step 4, load W_ih from .pt
     load_paths: str = ""
     log_jsonl: str = ""             # structured metrics sink
@@ -84,6 +93,8 @@ class G2VecConfig:
                 f"W_ih gather storage only; the fast path computes in fp32 "
                 f"regardless. Use --trainer-path general with a reduced "
                 f"dtype, or drop --dtype.")
+        if self.train_ckpt_every < 1:
+            raise ValueError("train_ckpt_every must be >= 1")
         if self.earlystop_every < 1:
             raise ValueError("earlystop_every must be >= 1")
         if self.epochs < 1:

@@ -729,16 +729,68 @@ class CbowTrainer:
                 return hist, -1, st.W, st.who, acc_tr
 
     # ------------------------------------------------------------------ train
+    # --------------------------------------------------- train-state ckpt
+    def save_train_state(self, st, path: str, extra: dict) -> None:
+        """Epoch-boundary training checkpoint: weights,
+        TF1-Adam moments, the fused-eval dlogit carry, early-stop
+        trackers, and a config fingerprint. Resuming from it continues
+        the EXACT trajectory (epoch bodies are deterministic)."""
+        blob = {
+            "W": st.W.cpu(), "who": st.who.cpu(),
+            "mW": st.mW.cpu(), "vW": st.vW.cpu(),
+            "mO": st.mO.cpu(), "vO": st.vO.cpu(),
+            "W_keep": st.W_keep.cpu(),
+            "dO_buf": (st.dO_buf.cpu() if st.dO_buf is not None else None),
+            "t_adam": st.t_adam, "epoch_idx": st.epoch_idx,
+            "fingerprint": {"n_genes": self.G, "hidden": self.h,
+                            "seed": self.cfg.seed, "lr": self.cfg.lr,
+                            "trainer_path": self.cfg.trainer_path,
+                            "world": self.ctx.world, "rank": self.ctx.rank},
+        }
+        blob.update(extra)
+        torch.save(blob, path)
+
+    def load_train_state(self, st, path: str) -> dict:
+        blob = torch.load(path, map_location="cpu", weights_only=False)
+        fp = blob["fingerprint"]
+        for key in ("n_genes", "hidden", "seed", "lr", "trainer_path"):
+            want = {"n_genes": self.G, "hidden": self.h,
+                    "seed": self.cfg.seed, "lr": self.cfg.lr,
+                    "trainer_path": self.cfg.trainer_path}[key]
+            if fp[key] != want:
+                raise ValueError(
+                    f"--resume-train checkpoint mismatch: {key} was "
+                    f"{fp[key]}, this run has {want}")
+        st.W.copy_(blob["W"].to(self.device))
+        st.who.copy_(blob["who"].to(self.device))
+        st.mW.copy_(blob["mW"].to(self.device))
+        st.vW.copy_(blob["vW"].to(self.device))
+        st.mO.copy_(blob["mO"].to(self.device))
+        st.vO.copy_(blob["vO"].to(self.device))
+        st.W_keep.copy_(blob["W_keep"].to(self.device))
+        if st.dO_buf is not None and blob["dO_buf"] is not None:
+            st.dO_buf.copy_(blob["dO_buf"].to(self.device))
+        st.t_adam = int(blob["t_adam"])
+        st.epoch_idx = int(blob["epoch_idx"])
+        if st.s_buf is not None:        # derived: recompute from W, who
+            ops.gemv_rows(st.W, st.who, st.s_buf)
+        return blob
+
     def train(self, ps: PathSet, pre_sharded: bool = False) -> TrainResult:
         cfg = self.cfg
         st = self.setup(ps, pre_sharded)
 
-        if (cfg.trainer_path != "general" and cfg.batch_size == 0 and
+        ckpting = bool(cfg.train_ckpt) or bool(cfg.resume_train)
+        if (not ckpting and
+                cfg.trainer_path != "general" and cfg.batch_size == 0 and
                 (self.device.type == "cuda" or
                  (cfg.early_stop and cfg.earlystop_every > 1))):
             # GPU fast path always; CPU too when --earlystop-every opts
             # into the k-granular runner (the flag must never be a
-            # silent no-op — cf. the round-1 --dtype finding)
+            # silent no-op — cf. the round-1 --dtype finding).
+            # Train-state checkpointing runs the synchronous loop on any
+            # device: checkpoints must be epoch-aligned, which the
+            # speculative pipeline is not.
             return self._train_pipelined(st)
 
         before_val, before_tr = -1.0, -1.0
@@ -746,13 +798,29 @@ class CbowTrainer:
         acc_hist: List[float] = []
         epoch_times: List[float] = []
         wall_to_acc = None
+        start_epoch = 0
+
+        def _rank_path(p: str) -> str:
+            # dO_buf / split shards are rank-local: world>1 checkpoints
+            # are one file per rank
+            return p if self.ctx.world == 1 else f"{p}.rank{self.ctx.rank}"
+
+        if cfg.resume_train:
+            blob = self.load_train_state(st, _rank_path(cfg.resume_train))
+            acc_hist = list(blob.get("acc_hist", []))
+            before_val = float(blob.get("before_val", -1.0))
+            before_tr = float(blob.get("before_tr", -1.0))
+            start_epoch = st.epoch_idx
+            self.log(f"    (resumed training state at epoch {start_epoch} "
+                     f"from {cfg.resume_train})")
         t0_all = time.perf_counter()
         display_step = 5
         blk_t0 = time.perf_counter()
 
         self.log("     Start training the modified CBOW with early stopping")
-        epochs_run = 0
-        for epoch in range(cfg.epochs):
+        epochs_run = start_epoch
+        acc_val, acc_tr = before_val, before_tr
+        for epoch in range(start_epoch, cfg.epochs):
             ep_t0 = time.perf_counter()
             acc_tr, acc_val = self.run_epoch(st)
             epochs_run = epoch + 1
@@ -775,6 +843,12 @@ class CbowTrainer:
                 break
             before_val, before_tr = acc_val, acc_tr
             st.W_keep.copy_(st.W)   # keep-last-good snapshot (G2Vec.py:283)
+            if (cfg.train_ckpt and
+                    (epoch + 1) % max(cfg.train_ckpt_every, 1) == 0):
+                self.save_train_state(st, _rank_path(cfg.train_ckpt),
+                                      {"acc_hist": acc_hist,
+                                       "before_val": before_val,
+                                       "before_tr": before_tr})
         self.log("    Optimization Finish")
 
         return TrainResult(W_ih=self._unrelabel(st.W_keep),

@@ -241,3 +241,69 @@ def test_gene_relabel_early_stop_unpermutes_kept_weights():
         outs[mode] = tr.train(ps)
     assert outs["on"].stop_epoch == outs["off"].stop_epoch
     assert torch.allclose(outs["on"].W_ih, outs["off"].W_ih, atol=1e-5)
+
+
+def test_train_state_checkpoint_resume_bitwise(tmp_path):
+    """--train-ckpt / --resume-train: resuming mid-training
+    continues the EXACT trajectory — epoch bodies are deterministic, so
+    the resumed run's weights and accuracy history are bitwise the
+    uninterrupted run's."""
+    ps = _random_pathset(G=60, P=240, seed=7)
+    def mk(**kw):
+        base = dict(hidden=64, epochs=20, early_stop=False, seed=3,
+                    device="cpu")
+        base.update(kw)
+        return G2VecConfig(**base)
+    full = CbowTrainer(mk(), ps.n_genes, torch.device("cpu"),
+                       log=lambda *a, **k: None).train(ps)
+
+    ck = str(tmp_path / "train_state.pt")
+    half = CbowTrainer(mk(epochs=10, train_ckpt=ck, train_ckpt_every=10),
+                       ps.n_genes, torch.device("cpu"),
+                       log=lambda *a, **k: None).train(ps)
+    assert half.epochs_run == 10
+    resumed = CbowTrainer(mk(resume_train=ck), ps.n_genes,
+                          torch.device("cpu"),
+                          log=lambda *a, **k: None).train(ps)
+    assert resumed.epochs_run == 20
+    assert resumed.acc_val_history == pytest.approx(full.acc_val_history,
+                                                    abs=0)
+    assert torch.equal(resumed.W_ih, full.W_ih)
+
+
+def test_train_state_resume_early_stop(tmp_path):
+    """Early stop after a resume reproduces the uninterrupted stop epoch
+    and keep-last-good weights."""
+    ps = _random_pathset(G=60, P=240, seed=9)
+    def mk(**kw):
+        base = dict(hidden=64, epochs=40, early_stop=True, seed=2,
+                    device="cpu")
+        base.update(kw)
+        return G2VecConfig(**base)
+    full = CbowTrainer(mk(), ps.n_genes, torch.device("cpu"),
+                       log=lambda *a, **k: None).train(ps)
+    assert full.stop_epoch >= 2, "fixture must stop after the ckpt point"
+    ck = str(tmp_path / "ts.pt")
+    CbowTrainer(mk(epochs=2, train_ckpt=ck, train_ckpt_every=2,
+                   early_stop=False),
+                ps.n_genes, torch.device("cpu"),
+                log=lambda *a, **k: None).train(ps)
+    resumed = CbowTrainer(mk(resume_train=ck), ps.n_genes,
+                          torch.device("cpu"),
+                          log=lambda *a, **k: None).train(ps)
+    assert resumed.stop_epoch == full.stop_epoch
+    assert torch.equal(resumed.W_ih, full.W_ih)
+
+
+def test_train_state_fingerprint_mismatch(tmp_path):
+    ps = _random_pathset(G=60, P=200, seed=1)
+    ck = str(tmp_path / "fp.pt")
+    CbowTrainer(G2VecConfig(hidden=64, epochs=5, early_stop=False, seed=3,
+                            device="cpu", train_ckpt=ck, train_ckpt_every=5),
+                ps.n_genes, torch.device("cpu"),
+                log=lambda *a, **k: None).train(ps)
+    bad = G2VecConfig(hidden=64, epochs=10, early_stop=False, seed=99,
+                      device="cpu", resume_train=ck)
+    with pytest.raises(ValueError, match="mismatch"):
+        CbowTrainer(bad, ps.n_genes, torch.device("cpu"),
+                    log=lambda *a, **k: None).train(ps)
