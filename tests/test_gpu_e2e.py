@@ -208,3 +208,39 @@ def test_rccl_allreduce_capturable_in_hipgraph():
         assert torch.allclose(t, torch.full_like(t, 8.0))
     finally:
         dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_train_state_resume_bitwise_on_gpu(tmp_path):
+    """Mid-training checkpoint/resume on GPU: exact continuation."""
+    import numpy as np
+    from g2vec_amd.models.cbow import CbowTrainer
+    from g2vec_amd.paths import PathSet
+    rng = np.random.default_rng(41)
+    G, P = 200, 400
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 15))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda", 0)
+    ps = PathSet(torch.tensor(genes, dtype=torch.int32, device=dev),
+                 torch.tensor(offs, dtype=torch.int32, device=dev),
+                 torch.tensor(labels, device=dev), G)
+
+    def mk(**kw):
+        base = dict(hidden=128, epochs=16, early_stop=False, seed=6,
+                    device="cuda")
+        base.update(kw)
+        return G2VecConfig(**base)
+
+    full = CbowTrainer(mk(), G, dev, log=lambda *a, **k: None).train(ps)
+    ck = str(tmp_path / "gs.pt")
+    CbowTrainer(mk(epochs=8, train_ckpt=ck, train_ckpt_every=8), G, dev,
+                log=lambda *a, **k: None).train(ps)
+    resumed = CbowTrainer(mk(resume_train=ck), G, dev,
+                          log=lambda *a, **k: None).train(ps)
+    assert resumed.acc_val_history[8:] == pytest.approx(
+        full.acc_val_history[8:], abs=0)
+    assert torch.equal(resumed.W_ih.cpu(), full.W_ih.cpu())
