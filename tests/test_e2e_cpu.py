@@ -244,3 +244,32 @@ def test_degenerate_inputs_fail_loudly(tiny_files, tmp_path):
                      torch.device("cpu"), log=lambda *a, **k: None)
     with pytest.raises(ValueError, match="no paths"):
         tr.train(ps)
+
+
+def test_failure_recovery_paths_cache_plus_train_resume(tiny_files, tmp_path):
+    """The full failure-recovery story: a run that
+    dies mid-training resumes from the step-3 path cache + the
+    mid-training state file and produces byte-identical outputs to an
+    uninterrupted run."""
+    paths_cache = str(tmp_path / "paths.pt")
+    ck = str(tmp_path / "state.pt")
+    base = dict(len_path=15, num_repetition=3, device="cpu", seed=0,
+                early_stop=False)
+
+    # uninterrupted 12-epoch run (also writes the path cache)
+    full = run(_cfg(tiny_files, tmp_path, epochs=12, save_paths=paths_cache,
+                    result_name=str(tmp_path / "full"), **base))
+
+    # "crashed" run: 6 epochs, checkpoint at 6
+    run(_cfg(tiny_files, tmp_path, epochs=6, load_paths=paths_cache,
+             train_ckpt=ck, train_ckpt_every=6,
+             result_name=str(tmp_path / "crash"), **base))
+    # recovery: same path cache + training state, continue to 12
+    rec = run(_cfg(tiny_files, tmp_path, epochs=12, load_paths=paths_cache,
+                   resume_train=ck,
+                   result_name=str(tmp_path / "rec"), **base))
+    assert rec["acc_val"] == pytest.approx(full["acc_val"], abs=0)
+    assert (tmp_path / "rec_vectors.txt").read_text() == \
+           (tmp_path / "full_vectors.txt").read_text()
+    assert (tmp_path / "rec_biomarkers.txt").read_text() == \
+           (tmp_path / "full_biomarkers.txt").read_text()
