@@ -36,20 +36,11 @@ def native():
 
 
 # ------------------------------------------------------------------ walks
-WALK_IMPL_DEFAULT = 0   # 0 = wave-per-walk; 1 = 16-lane subwave (4/wave).
-                        # Both emit bitwise-identical output; the default is
-                        # the A/B-measured winner (tools/bench_walks.py) and
-                        # G2VEC_WALK_IMPL overrides it for experiments.
-
-
 def random_walks(row_ptr, col_idx, weights, sources, num_repetition: int,
                  len_path: int, seed: int):
     if row_ptr.is_cuda:
-        import os
-        impl = int(os.environ.get("G2VEC_WALK_IMPL", WALK_IMPL_DEFAULT))
         return native().random_walks(row_ptr, col_idx, weights, sources,
-                                     num_repetition, len_path, seed,
-                                     impl=impl)
+                                     num_repetition, len_path, seed)
     return cpu_ref.random_walks(row_ptr, col_idx, weights, sources,
                                 num_repetition, len_path, seed)
 

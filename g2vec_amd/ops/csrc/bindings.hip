@@ -43,10 +43,7 @@ inline int grid_for(long long work_items, int per_block) {
 std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col_idx,
                                         torch::Tensor weights, torch::Tensor sources,
                                         int64_t num_repetition, int64_t len_path,
-                                        int64_t seed, int64_t impl) {
-  // impl: 0 = wave-per-walk (walk_kernel), 1 = 16-lane subwave
-  // (walk16_kernel — 4 walks/wave, 4x the latency chains in flight;
-  // bitwise-identical output). The ops layer picks the measured default.
+                                        int64_t seed) {
   CHECK_DEV(row_ptr); CHECK_CONT(row_ptr); CHECK_I32(row_ptr);
   CHECK_DEV(col_idx); CHECK_CONT(col_idx); CHECK_I32(col_idx);
   CHECK_DEV(weights); CHECK_CONT(weights); CHECK_F32(weights);
@@ -62,20 +59,6 @@ std::vector<torch::Tensor> random_walks(torch::Tensor row_ptr, torch::Tensor col
   if (n_walks == 0) return {nodes, lengths, hashes};
   int tsize = 256;                   // LDS hash set: >= 2x path length, pow2
   while (tsize < 2 * (int)len_path) tsize <<= 1;
-  if (impl == 1) {
-    const int spb = 16;              // 256 threads = 16 walk slots
-    const size_t lds = (size_t)spb * (len_path + tsize) * sizeof(int);
-    hipLaunchKernelGGL(walk16_kernel, dim3(grid_for(n_walks, spb)), dim3(256),
-                       lds, cur_stream(), row_ptr.data_ptr<int>(),
-                       col_idx.data_ptr<int>(), weights.data_ptr<float>(),
-                       sources.data_ptr<int>(),
-                       (int)n_src, n_walks, (int)num_repetition,
-                       (int)len_path, tsize, (uint64_t)seed,
-                       nodes.data_ptr<int>(), lengths.data_ptr<int>(),
-                       (long long*)hashes.data_ptr<int64_t>());
-    LAUNCH_CHECK();
-    return {nodes, lengths, hashes};
-  }
   const int wpb = 4;                 // 256 threads = 4 waves
   const size_t lds = (size_t)wpb * (len_path + tsize) * sizeof(int);
   hipLaunchKernelGGL(walk_kernel, dim3(grid_for(n_walks, wpb)), dim3(256), lds,
@@ -561,11 +544,7 @@ parse_expression_tsv(const std::string& path) {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("random_walks", &random_walks,
-        py::arg("row_ptr"), py::arg("col_idx"), py::arg("weights"),
-        py::arg("sources"), py::arg("num_repetition"), py::arg("len_path"),
-        py::arg("seed"), py::arg("impl") = 0,
-        "CSR biased random walks (gfx950); impl 0=wave/walk, 1=16-lane subwave");
+  m.def("random_walks", &random_walks, "CSR biased random walks (gfx950)");
   m.def("cbow_fwd_scalar", &cbow_fwd_scalar, "scalar CBOW forward + loss");
   m.def("scatter_dO_det", &scatter_dO_det, "deterministic c = X^T dO");
   m.def("scatter_dO_det_", &scatter_dO_det_,

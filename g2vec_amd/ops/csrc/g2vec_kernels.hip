@@ -965,210 +965,3 @@ eval_finish_kernel(const float* __restrict__ piece, const int* __restrict__ offs
     partials[2 * blockIdx.x + 1] = sm[4] + sm[5] + sm[6] + sm[7];
   }
 }
-
-// ========================================== K10 variant: 16-lane subwave walks
-// walk_kernel dedicates a 64-lane wave to ONE walk: at the ex_* graphs'
-// mean out-degree ~16 that leaves 3/4 of the lanes idle, and the
-// per-step chain (row_ptr -> row -> LDS probe -> scan -> pick, two
-// dependent L2 round-trips) is what bounds the kernel. This variant
-// walks FOUR paths per wave on 16-lane subgroups: 4x the independent
-// latency chains in flight per SIMD at ~VGPR-34 occupancy, same
-// semantics and the same per-(source,repetition) RNG stream — bitwise
-// identical output to walk_kernel / the CPU oracle. Rows with more
-// than 4*16 neighbors fall back to a 16-wide chunked loop (rare: the
-// real ex graphs top out at deg 269).
-#define SUBW16 16
-
-__device__ __forceinline__ float sub16_incl_scan(float v, int l16) {
-#pragma unroll
-  for (int o = 1; o < SUBW16; o <<= 1) {
-    const float t = __shfl_up(v, o);
-    if (l16 >= o) v += t;        // guard keeps the scan inside the group
-  }
-  return v;
-}
-
-extern "C" __global__ void __launch_bounds__(256)
-walk16_kernel(const int* __restrict__ row_ptr, const int* __restrict__ col_idx,
-              const float* __restrict__ wgt, const int* __restrict__ sources,
-              int n_src, long long n_walks, int num_rep, int len_path,
-              int tsize, uint64_t seed, int* __restrict__ out_nodes,
-              int* __restrict__ out_len, long long* __restrict__ out_hash) {
-  extern __shared__ int smem[];
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int l16 = threadIdx.x & (SUBW16 - 1);
-  const int slot = threadIdx.x >> 4;           // walk slot in block (0..15)
-  const int spb = blockDim.x >> 4;             // slots per block
-  const int gsh = (lane >> 4) << 4;            // subgroup base lane in wave
-  int* vis = smem + slot * len_path;
-  uint32_t* tab = (uint32_t*)(smem + spb * len_path) + slot * tsize;
-  const uint32_t tmask = (uint32_t)tsize - 1;
-  // the wave's four slots iterate in lockstep (shuffles span the wave)
-  const int wave_slot0 = slot & ~3;
-
-  for (long long wbase = (long long)blockIdx.x * spb + wave_slot0;
-       wbase < n_walks; wbase += (long long)gridDim.x * spb) {
-    const long long walk = wbase + (slot & 3);
-    const bool act = walk < n_walks;
-    int cur = 0, plen = 0;
-    uint64_t state = 0, hash = 0;
-    bool alive = act;
-    if (act) {
-      const int src = sources[walk % n_src];
-      const int rep = (int)(walk / n_src);
-      const uint64_t gid = (uint64_t)src * (uint64_t)num_rep + (uint64_t)rep;
-      state = seed ^ (uint64_t)(gid * 0x94D049BB133111EBULL + 1ULL);
-      (void)sm64_next(state);                  // warm draw (oracle parity)
-      cur = src;
-    }
-    for (int i = l16; i < tsize; i += SUBW16) tab[i] = HSET_EMPTY;
-
-    for (int step = 0; step < len_path; ++step) {
-      if (!__ballot(alive)) break;
-      int deg = 0, s = 0;
-      if (alive) {
-        if (l16 == 0) {
-          vis[plen] = cur;
-          hset_insert(tab, tmask, (uint32_t)cur);
-        }
-        ++plen;
-        hash += gene_hash_dev((uint32_t)cur);
-        s = row_ptr[cur];
-        deg = row_ptr[cur + 1] - s;
-        if (deg <= 0) alive = false;           // dead end: no draw
-      }
-      if (!__ballot(alive)) break;
-      if (alive) {
-        if (deg <= 4 * SUBW16) {
-          // register path: whole row in <= 4 chunks of 16
-          const int nchunk = (deg + SUBW16 - 1) >> 4;
-          float wreg[4];
-          int creg[4];
-          float partial = 0.f;
-#pragma unroll
-          for (int k = 0; k < 4; ++k) {
-            wreg[k] = 0.f;
-            creg[k] = -1;
-            const int j = (k << 4) + l16;
-            if (k < nchunk && j < deg) {
-              const int cc = col_idx[s + j];
-              creg[k] = cc;
-              float w = wgt[s + j];
-              if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
-              wreg[k] = w;
-              partial += w;
-            }
-          }
-          float tsum = partial;
-#pragma unroll
-          for (int o = 8; o; o >>= 1) tsum += __shfl_xor(tsum, o);
-          const float tot = tsum;              // same value on all 16 lanes
-          const uint64_t r = sm64_next(state); // drawn even on dead end
-          if (!(tot > 0.f)) { alive = false; }
-          else {
-            const float target = (float)(u01_from(r) * (double)tot);
-            int chosen = -1;
-            float base = 0.f;
-            unsigned int any_pos_last = 0;
-            int last_pos_cand = -1;
-            for (int k = 0; k < nchunk && chosen < 0; ++k) {
-              const float w = wreg[k];
-              const float scan = sub16_incl_scan(w, l16);
-              const float chunk_tot = __shfl(scan, gsh + SUBW16 - 1);
-              const bool hit = (w > 0.f) && (base + scan > target) &&
-                               (base + scan - w <= target);
-              const unsigned int mh =
-                  (unsigned int)((__ballot(hit) >> gsh) & 0xFFFFULL);
-              if (mh != 0u) {
-                chosen = __shfl(creg[k], gsh + (__ffs(mh) - 1));
-              } else {
-                const unsigned int mp =
-                    (unsigned int)((__ballot(w > 0.f) >> gsh) & 0xFFFFULL);
-                if (mp != 0u) {
-                  any_pos_last = 1;
-                  last_pos_cand = __shfl(creg[k], gsh + (31 - __clz(mp)));
-                }
-                base += chunk_tot;
-              }
-            }
-            if (chosen < 0) {
-              // rounding tail: last positive-weight (unvisited) neighbor
-              if (!any_pos_last) alive = false;
-              else chosen = last_pos_cand;
-            }
-            if (alive) cur = chosen;
-          }
-        } else {
-          // chunked fallback for rows wider than 64 neighbors
-          float partial = 0.f;
-          for (int j = l16; j < deg; j += SUBW16) {
-            const int cc = col_idx[s + j];
-            float w = wgt[s + j];
-            if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
-            partial += w;
-          }
-          float tsum = partial;
-#pragma unroll
-          for (int o = 8; o; o >>= 1) tsum += __shfl_xor(tsum, o);
-          const float tot = tsum;
-          const uint64_t r = sm64_next(state);
-          if (!(tot > 0.f)) { alive = false; }
-          else {
-            const float target = (float)(u01_from(r) * (double)tot);
-            int chosen = -1;
-            float base = 0.f;
-            for (int j0 = 0; j0 < deg && chosen < 0; j0 += SUBW16) {
-              const int j = j0 + l16;
-              float w = 0.f;
-              int cc = -1;
-              if (j < deg) {
-                cc = col_idx[s + j];
-                w = wgt[s + j];
-                if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
-              }
-              const float scan = sub16_incl_scan(w, l16);
-              const float chunk_tot = __shfl(scan, gsh + SUBW16 - 1);
-              const bool hit = (w > 0.f) && (base + scan > target) &&
-                               (base + scan - w <= target);
-              const unsigned int mh =
-                  (unsigned int)((__ballot(hit) >> gsh) & 0xFFFFULL);
-              if (mh != 0u) chosen = __shfl(cc, gsh + (__ffs(mh) - 1));
-              else base += chunk_tot;
-            }
-            if (chosen < 0) {
-              // backward sweep for the rounding tail
-              for (int j0 = ((deg - 1) >> 4) << 4; j0 >= 0 && chosen < 0;
-                   j0 -= SUBW16) {
-                const int j = j0 + l16;
-                float w = 0.f;
-                int cc = -1;
-                if (j < deg) {
-                  cc = col_idx[s + j];
-                  w = wgt[s + j];
-                  if (hset_contains(tab, tmask, (uint32_t)cc)) w = 0.f;
-                }
-                const unsigned int mp =
-                    (unsigned int)((__ballot(w > 0.f) >> gsh) & 0xFFFFULL);
-                if (mp != 0u) chosen = __shfl(cc, gsh + (31 - __clz(mp)));
-              }
-              if (chosen < 0) alive = false;   // cannot happen when tot > 0
-            }
-            if (alive) cur = chosen;
-          }
-        }
-      }
-    }
-
-    if (act) {
-      const long long outb = walk * (long long)len_path;
-      for (int k = l16; k < len_path; k += SUBW16)
-        out_nodes[outb + k] = (k < plen) ? vis[k] : -1;
-      if (l16 == 0) {
-        out_len[walk] = plen;
-        out_hash[walk] = (long long)hash;
-      }
-    }
-    // reset the hash table for the next grid-stride walk
-    for (int i = l16; i < tsize; i += SUBW16) tab[i] = HSET_EMPTY;
-  }
-}

@@ -35,9 +35,6 @@ def main():
     for g in (0, 1):
         print(f"group {g}: nnz {graphs[g].col_idx.numel()}", file=sys.stderr)
 
-    import os
-    impl = os.environ.get("G2VEC_WALK_IMPL", "?")
-    print(f"walk impl = {impl}", file=sys.stderr)
     # warmup
     for g in (0, 1):
         generate_walks(graphs[g], args.len_path, args.reps, 0, g)
