@@ -44,6 +44,9 @@ def build_parser() -> argparse.ArgumentParser:
                    help="|PCC| cutoff for graph edges (reference: 0.5)")
     p.add_argument("--kmeans", choices=["auto", "sklearn", "torch"],
                    default="auto", help="L-group clustering backend")
+    p.add_argument("--activation", choices=["none", "relu"], default="none",
+                   help="hidden activation (general chain only; the fast "
+                        "path exploits the reference net's linearity)")
     p.add_argument("--trainer-path", choices=["fast", "general"], default="fast")
     p.add_argument("--batch-size", type=int, default=0,
                    help="0 = full batch (reference semantics)")
@@ -86,6 +89,7 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         seed=(None if a.seed < 0 else a.seed), dtype=a.dtype, device=a.device,
         pcc_mode=a.pcc_mode, pcc_threshold=a.pcc_threshold,
         kmeans_backend=a.kmeans, trainer_path=a.trainer_path,
+        activation=a.activation,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, earlystop_every=a.earlystop_every,
         train_ckpt=a.train_ckpt, train_ckpt_every=a.train_ckpt_every,

@@ -52,6 +52,12 @@ class G2VecConfig:
     batch_size: int = 0             # 0 = full batch (reference semantics, G2Vec.py:264)
     trainer_path: str = "fast"      # "fast": collapsed rank-1 path (linear-net algebra)
                                     # "general": full gather/scatter kernel chain (K1-K8)
+    activation: str = "none"        # hidden activation: "none" (reference
+                                    # semantics, G2Vec.py:238-240) | "relu"
+                                    # (opt-in non-linear successor on the
+                                    # GENERAL chain; the fast path's linear
+                                    # collapse does not apply, so validate()
+                                    # requires --trainer-path general)
     save_paths: str = ""            # cache generated path set (de-facto checkpoint)
     save_model: str = ""            # save trained W_ih/W_ho + metadata (.pt)
     train_ckpt: str = ""            # mid-TRAINING checkpoint file: weights +
@@ -109,6 +115,13 @@ class G2VecConfig:
             raise ValueError(f"bad pcc_mode {self.pcc_mode}")
         if self.trainer_path not in ("fast", "general"):
             raise ValueError(f"bad trainer_path {self.trainer_path}")
+        if self.activation not in ("none", "relu"):
+            raise ValueError(f"bad activation {self.activation}")
+        if self.activation != "none" and self.trainer_path != "general":
+            raise ValueError(
+                "activation != none requires --trainer-path general: the "
+                "fast path exploits the reference net's LINEARITY "
+                "(o = sum s_g), which a hidden activation breaks")
         if self.kmeans_backend not in ("auto", "sklearn", "torch"):
             raise ValueError(f"bad kmeans_backend {self.kmeans_backend}")
         if self.gene_relabel not in ("auto", "on", "off"):

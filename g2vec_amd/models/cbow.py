@@ -946,12 +946,14 @@ class CbowTrainer:
         mW, vW, mO, vO = st.mW, st.vW, st.mO, st.vO
         genes, offsets, labels = self._slice(tr, lo, hi)
         Wg = W16 if W16 is not None else W
+        act = 1 if self.cfg.activation == "relu" else 0
         _loss, _corr, dO, H = ops.cbow_fwd(Wg, who, genes, offsets, labels,
-                                           inv_b, True)
+                                           inv_b, True, act=act)
         full = lo == 0 and hi == tr.n_paths
         dW = ops.cbow_bwd_rows(who, genes, offsets, dO, self.G,
-                               plan=(st.plan if full else None))
-        grad_who = torch.mv(H.t(), dO)
+                               plan=(st.plan if full else None),
+                               H_pre=(H if act else None))
+        grad_who = torch.mv((torch.relu(H) if act else H).t(), dO)
         self.ctx.allreduce_(dW)
         self.ctx.allreduce_(grad_who)
         ops.adam_dense(W, mW, vW, dW, t, self.cfg.lr, self.B1, self.B2, self.EPS)
@@ -965,8 +967,9 @@ class CbowTrainer:
             correct = torch.zeros((), dtype=torch.float32, device=self.device)
         elif self.cfg.trainer_path == "general":
             Wg = W16 if W16 is not None else W
-            _l, corr, _d, _h = ops.cbow_fwd(Wg, who, ps.genes, ps.offsets,
-                                            ps.labels, 1.0, False)
+            _l, corr, _d, _h = ops.cbow_fwd(
+                Wg, who, ps.genes, ps.offsets, ps.labels, 1.0, False,
+                act=1 if self.cfg.activation == "relu" else 0)
             correct = corr.sum()
         else:
             s = torch.mv(W, who)

@@ -202,27 +202,36 @@ def adam_dense(W, m, v, grad, t: int, lr: float, b1: float, b2: float,
 
 
 # ------------------------------------------------------------------ CBOW general (kernel-chain) path
-def cbow_fwd(W, who, genes, offsets, labels, inv_b: float, want_grad: bool):
+def cbow_fwd(W, who, genes, offsets, labels, inv_b: float, want_grad: bool,
+             act: int = 0):
+    """act: 0 linear (reference semantics); 1 ReLU on the hidden vector
+    (opt-in non-linear successor; returned H is the pre-activation)."""
     if W.is_cuda:
         loss, correct, dO, H = native().cbow_fwd(
-            W, who, genes, offsets, labels, float(inv_b), bool(want_grad))
+            W, who, genes, offsets, labels, float(inv_b), bool(want_grad),
+            act=int(act))
         if not want_grad:
             return loss, correct, None, None
         return loss, correct, dO, H
-    return cpu_ref.cbow_fwd(W, who, genes, offsets, labels, inv_b, want_grad)
+    return cpu_ref.cbow_fwd(W, who, genes, offsets, labels, inv_b, want_grad,
+                            act=act)
 
 
 def cbow_bwd_rows(who, genes, offsets, dO, n_genes: int,
-                  plan: Optional[ScatterPlan] = None):
+                  plan: Optional[ScatterPlan] = None,
+                  H_pre: Optional[torch.Tensor] = None):
     """General-path dW_ih backward. On GPU this is the deterministic
     per-gene-segment kernel (no atomics) driven by the same ScatterPlan as
-    the fast path's c-reduction."""
+    the fast path's c-reduction. H_pre given = ReLU backward mask source
+    (the forward's pre-activation H)."""
     if dO.is_cuda:
         if plan is None:
             plan = build_scatter_plan(genes, offsets, n_genes)
         return native().cbow_bwd_rows(who, plan.inst_path, plan.seg_start,
-                                      plan.seg_gene, dO, int(n_genes))
-    return cpu_ref.cbow_bwd_rows(who, genes, offsets, dO, n_genes)
+                                      plan.seg_gene, dO, int(n_genes),
+                                      Hpre=H_pre)
+    return cpu_ref.cbow_bwd_rows(who, genes, offsets, dO, n_genes,
+                                 H_pre=H_pre)
 
 
 def gemv_rows(W, x, out) -> None:

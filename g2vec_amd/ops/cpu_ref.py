@@ -204,17 +204,18 @@ def adam_dense(W: torch.Tensor, m: torch.Tensor, v: torch.Tensor,
 # ------------------------------------------------------------------ CBOW (general kernel-chain path)
 def cbow_fwd(W: torch.Tensor, who: torch.Tensor, genes: torch.Tensor,
              offsets: torch.Tensor, labels: torch.Tensor, inv_b: float,
-             want_grad: bool):
+             want_grad: bool, act: int = 0):
     """Row-gather forward: H_p = sum of W rows of the path's genes (sum, not
-    mean — X is 0/1, G2Vec.py:239), o = H @ who. Returns
-    (loss, correct, dO|None, H|None)."""
+    mean — X is 0/1, G2Vec.py:239), o = act(H) @ who. act: 0 linear
+    (reference), 1 ReLU (opt-in non-linear successor). Returns
+    (loss, correct, dO|None, H|None) — H is the PRE-activation."""
     Wf = W.float()
     counts = (offsets[1:] - offsets[:-1]).long()
     P = len(counts)
     seg = torch.repeat_interleave(torch.arange(P), counts)
     H = torch.zeros(P, Wf.shape[1], dtype=torch.float32)
     H.index_add_(0, seg, Wf[genes.long()])
-    o = H @ who
+    o = (torch.relu(H) if act else H) @ who
     y = labels.float()
     loss = torch.clamp(o, min=0) - o * y + torch.log1p(torch.exp(-o.abs()))
     correct = ((o > 0).float() == y).float()
@@ -225,11 +226,15 @@ def cbow_fwd(W: torch.Tensor, who: torch.Tensor, genes: torch.Tensor,
 
 
 def cbow_bwd_rows(who: torch.Tensor, genes: torch.Tensor, offsets: torch.Tensor,
-                  dO: torch.Tensor, n_genes: int) -> torch.Tensor:
-    """dW_ih = X^T (dO who^T): scatter-add dH rows into the touched gene rows."""
+                  dO: torch.Tensor, n_genes: int,
+                  H_pre: torch.Tensor = None) -> torch.Tensor:
+    """dW_ih = X^T (dO who^T [(.) relu_mask]): scatter-add dH rows into the
+    touched gene rows. H_pre given = ReLU backward (mask = H_pre > 0)."""
     counts = (offsets[1:] - offsets[:-1]).long()
     seg = torch.repeat_interleave(torch.arange(len(counts)), counts)
     dH = dO[:, None] * who[None, :]
+    if H_pre is not None:
+        dH = dH * (H_pre > 0).float()
     dW = torch.zeros(n_genes, who.shape[0], dtype=torch.float32)
     dW.index_add_(0, genes.long(), dH[seg])
     return dW

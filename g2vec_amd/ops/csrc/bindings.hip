@@ -263,7 +263,7 @@ void adam_dense(torch::Tensor W, torch::Tensor m, torch::Tensor v,
 std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
                                     torch::Tensor genes, torch::Tensor offs,
                                     torch::Tensor labels, double inv_b,
-                                    bool want_grad) {
+                                    bool want_grad, int64_t act) {
   CHECK_DEV(W); CHECK_CONT(W);
   CHECK_DEV(who); CHECK_CONT(who); CHECK_F32(who);
   CHECK_DEV(genes); CHECK_CONT(genes); CHECK_I32(genes);
@@ -293,7 +293,8 @@ std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
                      cur_stream(), PTR, who.data_ptr<float>(),                \
                      genes.data_ptr<int>(), offs.data_ptr<int>(),             \
                      labels.data_ptr<float>(), P, (float)inv_b, h, Hp,        \
-                     loss.data_ptr<float>(), correct.data_ptr<float>(), dOp)
+                     loss.data_ptr<float>(), correct.data_ptr<float>(), dOp,  \
+                     (int)act)
   if (bf16) {
     const bf16_bits* Wp = (const bf16_bits*)W.data_ptr<at::BFloat16>();
     switch (hpl) {
@@ -329,12 +330,15 @@ std::vector<torch::Tensor> cbow_fwd(torch::Tensor W, torch::Tensor who,
 
 torch::Tensor cbow_bwd_rows(torch::Tensor who, torch::Tensor inst_path,
                             torch::Tensor seg_start, torch::Tensor seg_gene,
-                            torch::Tensor dO, int64_t n_genes) {
+                            torch::Tensor dO, int64_t n_genes,
+                            std::optional<torch::Tensor> Hpre) {
+  // Hpre: pre-activation H for the ReLU backward mask (absent = linear)
   CHECK_DEV(who); CHECK_CONT(who); CHECK_F32(who);
   CHECK_DEV(inst_path); CHECK_CONT(inst_path); CHECK_I32(inst_path);
   CHECK_DEV(seg_start); CHECK_CONT(seg_start); CHECK_I32(seg_start);
   CHECK_DEV(seg_gene); CHECK_CONT(seg_gene); CHECK_I32(seg_gene);
   CHECK_DEV(dO); CHECK_CONT(dO); CHECK_F32(dO);
+  if (Hpre) { CHECK_DEV(*Hpre); CHECK_CONT(*Hpre); CHECK_F32(*Hpre); }
   const int h = (int)who.numel();
   const int hpl = h / 64;
   TORCH_CHECK(h % 64 == 0 && hpl >= 1 && hpl <= 16 &&
@@ -349,7 +353,8 @@ torch::Tensor cbow_bwd_rows(torch::Tensor who, torch::Tensor inst_path,
                      0, cur_stream(), who.data_ptr<float>(),                  \
                      inst_path.data_ptr<int>(), seg_start.data_ptr<int>(),    \
                      seg_gene.data_ptr<int>(), n_seg,                         \
-                     dO.data_ptr<float>(), h, dW.data_ptr<float>())
+                     dO.data_ptr<float>(), h, dW.data_ptr<float>(),           \
+                     Hpre ? Hpre->data_ptr<float>() : nullptr)
   switch (hpl) {
     case 1: BWD_CASE(1); break;
     case 2: BWD_CASE(2); break;
@@ -562,8 +567,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused train/val correct-count eval (in-place counts[2])");
   m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");
   m.def("adam_dense", &adam_dense, "TF1 Adam, dense grad");
-  m.def("cbow_fwd", &cbow_fwd, "row-gather CBOW forward");
-  m.def("cbow_bwd_rows", &cbow_bwd_rows, "scatter-add CBOW backward");
+  m.def("cbow_fwd", &cbow_fwd,
+        py::arg("W"), py::arg("who"), py::arg("genes"), py::arg("offs"),
+        py::arg("labels"), py::arg("inv_b"), py::arg("want_grad"),
+        py::arg("act") = 0,
+        "row-gather CBOW forward (act: 0 linear, 1 ReLU)");
+  m.def("cbow_bwd_rows", &cbow_bwd_rows,
+        py::arg("who"), py::arg("inst_path"), py::arg("seg_start"),
+        py::arg("seg_gene"), py::arg("dO"), py::arg("n_genes"),
+        py::arg("Hpre") = py::none(),
+        "deterministic scatter CBOW backward (Hpre = ReLU mask source)");
   m.def("pcc_edges", &pcc_edges, "per-edge |PCC|");
   m.def("corr_gemm", &corr_gemm, "MFMA f32 correlation GEMM");
   m.def("trunc_normal_", &trunc_normal_,

@@ -555,7 +555,11 @@ cbow_fwd_kernel(const WT* __restrict__ W, const float* __restrict__ who,
                 const int* __restrict__ genes, const int* __restrict__ offs,
                 const float* __restrict__ labels, long long P, float inv_b,
                 int h, float* __restrict__ H, float* __restrict__ loss,
-                float* __restrict__ correct, float* __restrict__ dO) {
+                float* __restrict__ correct, float* __restrict__ dO,
+                int act) {
+  // act: 0 = linear (reference semantics, G2Vec.py:238-240); 1 = ReLU on
+  // the hidden vector (opt-in non-linear successor; H stores the
+  // PRE-activation so the backward recovers the mask)
   const int lane = threadIdx.x & (WAVE - 1);
   const int wib = threadIdx.x >> 6;
   const int wpb = blockDim.x >> 6;
@@ -579,7 +583,8 @@ cbow_fwd_kernel(const WT* __restrict__ W, const float* __restrict__ who,
     float wv[HPL];
     __builtin_memcpy(wv, who + col0, HPL * sizeof(float));
 #pragma unroll
-    for (int k = 0; k < HPL; ++k) op += acc[k] * wv[k];
+    for (int k = 0; k < HPL; ++k)
+      op += (act ? fmaxf(acc[k], 0.f) : acc[k]) * wv[k];
     const float o = wave_sum(op);
     if (H) {
       float* hrow = H + p * (long long)h + col0;
@@ -609,7 +614,10 @@ cbow_bwd_rows_det_kernel(const float* __restrict__ who,
                          const int* __restrict__ seg_start,
                          const int* __restrict__ seg_gene, int n_seg,
                          const float* __restrict__ dO, int h,
-                         float* __restrict__ dW) {
+                         float* __restrict__ dW,
+                         const float* __restrict__ Hpre) {
+  // Hpre != nullptr: ReLU backward — dH_p = dO_p * who (.) 1[Hpre_p > 0]
+  // (one extra wide load per instance; linear path passes nullptr)
   const int lane = threadIdx.x & (WAVE - 1);
   const int wib = threadIdx.x >> 6;
   const int wpb = blockDim.x >> 6;
@@ -622,10 +630,23 @@ cbow_bwd_rows_det_kernel(const float* __restrict__ who,
     float acc[HPL];
 #pragma unroll
     for (int k = 0; k < HPL; ++k) acc[k] = 0.f;
-    for (int i = lo; i < hi; ++i) {
-      const float g = dO[inst_path[i]];       // general shape: dH row source
+    if (Hpre) {
+      for (int i = lo; i < hi; ++i) {
+        const long long pi = inst_path[i];
+        const float g = dO[pi];
+        float hrow[HPL];
+        __builtin_memcpy(hrow, Hpre + pi * (long long)h + col0,
+                         HPL * sizeof(float));
 #pragma unroll
-      for (int k = 0; k < HPL; ++k) acc[k] += g * wv[k];
+        for (int k = 0; k < HPL; ++k)
+          acc[k] += (hrow[k] > 0.f) ? g * wv[k] : 0.f;
+      }
+    } else {
+      for (int i = lo; i < hi; ++i) {
+        const float g = dO[inst_path[i]];     // general shape: dH row source
+#pragma unroll
+        for (int k = 0; k < HPL; ++k) acc[k] += g * wv[k];
+      }
     }
     float* row = dW + (long long)seg_gene[s] * h + col0;
     __builtin_memcpy(row, acc, HPL * sizeof(float));
@@ -800,7 +821,7 @@ f32_to_bf16_kernel(const float* __restrict__ in, uint16_t* __restrict__ out,
 #define INSTANTIATE_FWD(WT, HPL)                                              \
   template __global__ void cbow_fwd_kernel<WT, HPL>(                          \
       const WT*, const float*, const int*, const int*, const float*,          \
-      long long, float, int, float*, float*, float*, float*);
+      long long, float, int, float*, float*, float*, float*, int);
 INSTANTIATE_FWD(float, 1)
 INSTANTIATE_FWD(float, 2)
 INSTANTIATE_FWD(float, 4)
@@ -820,7 +841,7 @@ INSTANTIATE_FWD(fp16_bits, 16)
 #define INSTANTIATE_BWD(HPL)                                                  \
   template __global__ void cbow_bwd_rows_det_kernel<HPL>(                     \
       const float*, const int*, const int*, const int*, int, const float*,    \
-      int, float*);
+      int, float*, const float*);
 INSTANTIATE_BWD(1)
 INSTANTIATE_BWD(2)
 INSTANTIATE_BWD(4)

@@ -330,3 +330,39 @@ def test_eval_scan_matches_subwave_kernel_and_oracle():
     assert float(counts_b[0]) == float(corr[:p_split].sum())
     assert float(counts_b[1]) == float(corr[p_split:].sum())
     assert torch.allclose(dO_b.cpu(), d[:p_split], atol=1e-5)
+
+
+@pytest.mark.timeout(600)
+def test_relu_general_kernels_match_oracle_on_gpu():
+    """--activation relu GPU kernels (fwd act + masked backward) vs the
+    CPU oracle."""
+    import numpy as np
+
+    from g2vec_amd.ops import cpu_ref
+    rng = np.random.default_rng(19)
+    G, P, h = 300, 500, 128
+    genes, offs, labels = [], [0], []
+    for _ in range(P):
+        L = int(rng.integers(1, 20))
+        genes += rng.choice(G, size=L, replace=False).tolist()
+        offs.append(offs[-1] + L)
+        labels.append(float(rng.integers(0, 2)))
+    dev = torch.device("cuda")
+    W = torch.randn(G, h, device=dev) * 0.1
+    who = torch.randn(h, device=dev) * 0.1
+    g_t = torch.tensor(genes, dtype=torch.int32, device=dev)
+    o_t = torch.tensor(offs, dtype=torch.int32, device=dev)
+    l_t = torch.tensor(labels, dtype=torch.float32, device=dev)
+    inv_b = 1.0 / P
+    loss_g, corr_g, dO_g, H_g = ops.cbow_fwd(W, who, g_t, o_t, l_t, inv_b,
+                                             True, act=1)
+    dW_g = ops.cbow_bwd_rows(who, g_t, o_t, dO_g, G, H_pre=H_g)
+    loss_c, corr_c, dO_c, H_c = cpu_ref.cbow_fwd(
+        W.cpu(), who.cpu(), g_t.cpu(), o_t.cpu(), l_t.cpu(), inv_b, True,
+        act=1)
+    dW_c = cpu_ref.cbow_bwd_rows(who.cpu(), g_t.cpu(), o_t.cpu(), dO_c, G,
+                                 H_pre=H_c)
+    assert torch.allclose(loss_g.cpu(), loss_c, atol=1e-5)
+    assert torch.equal(corr_g.cpu(), corr_c)
+    assert torch.allclose(H_g.cpu(), H_c, atol=1e-5)
+    assert torch.allclose(dW_g.cpu(), dW_c, atol=1e-5)

@@ -307,3 +307,50 @@ def test_train_state_fingerprint_mismatch(tmp_path):
     with pytest.raises(ValueError, match="mismatch"):
         CbowTrainer(bad, ps.n_genes, torch.device("cpu"),
                     log=lambda *a, **k: None).train(ps)
+
+
+def test_relu_general_matches_dense_autograd():
+    """--activation relu on the general chain: gradients and trajectory
+    must match a dense autograd model o = relu(X W) who (the opt-in
+    non-linear successor the general path was built to host)."""
+    ps = _random_pathset(G=50, P=180, seed=4)
+    X = _dense_X(ps)
+    cfg = G2VecConfig(hidden=64, epochs=5, early_stop=False, seed=1,
+                      device="cpu", trainer_path="general",
+                      activation="relu")
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **k: None)
+    st = tr.setup(ps)
+    W0 = st.W.clone().requires_grad_(True)
+    who0 = st.who.clone().requires_grad_(True)
+
+    # one manual autograd step on the dense model over the SAME train split
+    tr_ps = st.tr
+    Xtr = _dense_X(tr_ps)
+    y = tr_ps.labels.float()
+    o = torch.relu(Xtr @ W0) @ who0
+    loss = torch.nn.functional.binary_cross_entropy_with_logits(
+        o, y, reduction="sum") / tr.n_tr_global
+    loss.backward()
+
+    from g2vec_amd import ops as _ops
+    act = 1
+    _l, _c, dO, H = _ops.cbow_fwd(st.W, st.who, tr_ps.genes, tr_ps.offsets,
+                                  tr_ps.labels, st.inv_b, True, act=act)
+    dW = _ops.cbow_bwd_rows(st.who, tr_ps.genes, tr_ps.offsets, dO,
+                            ps.n_genes, H_pre=H)
+    grad_who = torch.mv(torch.relu(H).t(), dO)
+    assert torch.allclose(dW, W0.grad, atol=1e-5)
+    assert torch.allclose(grad_who, who0.grad, atol=1e-5)
+
+    # the full training loop runs and yields a sane trajectory
+    res = tr.train(ps)
+    assert all(0.0 <= a <= 1.0 for a in res.acc_val_history)
+
+
+def test_activation_requires_general_path():
+    import pytest as _pt
+    with _pt.raises(ValueError, match="general"):
+        G2VecConfig(activation="relu").validate()
+    with _pt.raises(ValueError, match="activation"):
+        G2VecConfig(activation="gelu", trainer_path="general").validate()
