@@ -72,7 +72,6 @@ class G2VecConfig:
     load_model: str = ""            # resume: skip step 4, load W_ih from .pt
     load_paths: str = ""
     log_jsonl: str = ""             # structured metrics sink
-    deterministic_grads: bool = False  # bitwise-reproducible dW_ih reduction (no atomics)
     gene_relabel: str = "auto"      # trainer-internal gene-id relabeling for
                                     # gather locality: "on" | "off" | "auto"
                                     # (auto = on at n_genes >= 100k, where the
