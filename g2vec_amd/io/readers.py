@@ -35,12 +35,28 @@ def load_expression(path: str, use_native: bool = True) -> dict:
         samples = header[1:]
         genes: List[str] = []
         rows: List[np.ndarray] = []
-        for line in fin:
+        n_vals = None
+        for ln, line in enumerate(fin, start=2):
             parts = line.rstrip("\n").split("\t")
             if len(parts) < 2:
-                continue
+                continue            # lenient: blank/short rows skipped
+            try:
+                vals = np.asarray(parts[1:], dtype=np.float32)
+            except ValueError as e:
+                raise ValueError(
+                    f"{path}:{ln}: non-numeric expression value for gene "
+                    f"{parts[0]!r}: {e}") from None
+            if n_vals is None:
+                n_vals = len(vals)
+            elif len(vals) != n_vals:
+                raise ValueError(
+                    f"{path}:{ln}: gene {parts[0]!r} has {len(vals)} values, "
+                    f"expected {n_vals} (one per sample column)")
             genes.append(parts[0])
-            rows.append(np.asarray(parts[1:], dtype=np.float32))
+            rows.append(vals)
+    if not rows:
+        raise ValueError(f"{path}: no expression rows (empty or header-only "
+                         f"file)")
     expr = np.stack(rows, axis=0).T  # gene-wise -> sample-wise, like G2Vec.py:498
     return {"sample": np.array(samples), "expr": expr, "gene": np.array(genes)}
 
@@ -50,11 +66,16 @@ def load_clinical(path: str) -> Dict[str, int]:
     result: Dict[str, int] = {}
     with open(path) as fin:
         fin.readline()
-        for line in fin:
+        for ln, line in enumerate(fin, start=2):
             parts = line.rstrip().split("\t")
             if len(parts) < 2:
-                continue
-            result[parts[0]] = int(parts[1])
+                continue            # lenient: blank/short rows skipped
+            try:
+                result[parts[0]] = int(parts[1])
+            except ValueError:
+                raise ValueError(
+                    f"{path}:{ln}: clinical label for {parts[0]!r} must be "
+                    f"an integer (0 good / 1 poor), got {parts[1]!r}") from None
     return result
 
 

@@ -70,3 +70,27 @@ def test_write_vectors_engines_identical(tmp_path):
     a = write_vectors(str(tmp_path / "a"), mat, genes, engine="numpy")
     b = write_vectors(str(tmp_path / "b"), mat, genes, engine="pandas")
     assert open(a).read() == open(b).read()
+
+
+def test_malformed_inputs_raise_with_context(tmp_path):
+    """Parser errors name the file, line, and offending value (the
+    reference crashes with bare ValueErrors/IndexErrors)."""
+    import pytest
+
+    from g2vec_amd.io import readers
+    e1 = tmp_path / "ragged.tsv"
+    e1.write_text("PATIENT\tS1\tS2\nG1\t1.0\nG2\t1.0\t2.0\n")
+    with pytest.raises(ValueError, match=r"ragged\.tsv:2.*expected|values"):
+        readers.load_expression(str(e1), use_native=False)
+    e2 = tmp_path / "nonnum.tsv"
+    e2.write_text("PATIENT\tS1\nG1\tabc\n")
+    with pytest.raises(ValueError, match=r"nonnum\.tsv:2.*non-numeric"):
+        readers.load_expression(str(e2), use_native=False)
+    e3 = tmp_path / "empty.tsv"
+    e3.write_text("")
+    with pytest.raises(ValueError, match="no expression rows"):
+        readers.load_expression(str(e3), use_native=False)
+    c1 = tmp_path / "badlabel.tsv"
+    c1.write_text("H\tL\nS1\tx\n")
+    with pytest.raises(ValueError, match=r"badlabel\.tsv:2.*integer"):
+        readers.load_clinical(str(c1))
