@@ -35,7 +35,6 @@ def load_expression(path: str, use_native: bool = True) -> dict:
         samples = header[1:]
         genes: List[str] = []
         rows: List[np.ndarray] = []
-        n_vals = None
         for ln, line in enumerate(fin, start=2):
             parts = line.rstrip("\n").split("\t")
             if len(parts) < 2:
@@ -46,12 +45,10 @@ def load_expression(path: str, use_native: bool = True) -> dict:
                 raise ValueError(
                     f"{path}:{ln}: non-numeric expression value for gene "
                     f"{parts[0]!r}: {e}") from None
-            if n_vals is None:
-                n_vals = len(vals)
-            elif len(vals) != n_vals:
+            if len(vals) != len(samples):
                 raise ValueError(
                     f"{path}:{ln}: gene {parts[0]!r} has {len(vals)} values, "
-                    f"expected {n_vals} (one per sample column)")
+                    f"expected {len(samples)} (one per sample column)")
             genes.append(parts[0])
             rows.append(vals)
     if not rows:

@@ -525,19 +525,41 @@ parse_expression_tsv(const std::string& path) {
   std::vector<std::string> genes;
   std::vector<float> vals;
   vals.reserve(S * 4096);
+  long long ln = 1;
   while (std::getline(f, line)) {
+    ++ln;
     if (line.empty()) continue;
     const char* p = line.c_str();
     const char* tab = std::strchr(p, '\t');
-    if (!tab) continue;
+    if (!tab) continue;               // lenient: short rows skipped (like
+                                      // the Python fallback)
     genes.emplace_back(p, tab - p);
     const char* q = tab + 1;
     for (size_t i = 0; i < S; ++i) {
+      TORCH_CHECK(*q != '\0',
+                  path, ":", ln, ": gene '", genes.back(), "' has ", i,
+                  " values, expected ", S, " (one per sample column)");
       char* end = nullptr;
-      vals.push_back(std::strtof(q, &end));
-      q = (end && *end == '\t') ? end + 1 : end;
+      const float v = std::strtof(q, &end);
+      TORCH_CHECK(end != q && (*end == '\t' || *end == '\0' ||
+                               *end == '\r' || *end == '\n'),
+                  path, ":", ln, ": non-numeric expression value for gene '",
+                  genes.back(), "'");
+      vals.push_back(v);
+      const bool last = (i + 1 == S);
+      TORCH_CHECK(last || *end == '\t',
+                  path, ":", ln, ": gene '", genes.back(), "' has ", i + 1,
+                  " values, expected ", S, " (one per sample column)");
+      q = (*end == '\t') ? end + 1 : end;
     }
+    // trailing extra columns = ragged row
+    while (*q == '\r' || *q == '\n') ++q;
+    TORCH_CHECK(*q == '\0',
+                path, ":", ln, ": gene '", genes.back(),
+                "' has more than ", S, " values (one per sample column)");
   }
+  TORCH_CHECK(!genes.empty(), path,
+              ": no expression rows (empty or header-only file)");
   const size_t G = genes.size();
   auto t = torch::from_blob(vals.data(), {(long long)G, (long long)S},
                             torch::TensorOptions().dtype(at::kFloat))

@@ -80,8 +80,10 @@ def test_malformed_inputs_raise_with_context(tmp_path):
     from g2vec_amd.io import readers
     e1 = tmp_path / "ragged.tsv"
     e1.write_text("PATIENT\tS1\tS2\nG1\t1.0\nG2\t1.0\t2.0\n")
-    with pytest.raises(ValueError, match=r"ragged\.tsv:2.*expected|values"):
-        readers.load_expression(str(e1), use_native=False)
+    for native in (False, True):
+        with pytest.raises((ValueError, RuntimeError),
+                           match=r"ragged(\.tsv)?:2.*expected 2"):
+            readers.load_expression(str(e1), use_native=native)
     e2 = tmp_path / "nonnum.tsv"
     e2.write_text("PATIENT\tS1\nG1\tabc\n")
     with pytest.raises(ValueError, match=r"nonnum\.tsv:2.*non-numeric"):
