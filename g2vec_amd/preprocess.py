@@ -19,11 +19,20 @@ class SampleMismatchError(RuntimeError):
 
 def match_labels(clinical: Dict[str, int], samples: np.ndarray) -> np.ndarray:
     try:
-        return np.array([clinical[s] for s in samples], dtype=np.int64)
+        labels = np.array([clinical[s] for s in samples], dtype=np.int64)
     except KeyError as e:
         raise SampleMismatchError(
             f"sample {e.args[0]!r} present in expression data but missing from "
             f"clinical data (reference exits(1) here, G2Vec.py:432-433)") from e
+    bad = np.flatnonzero((labels != 0) & (labels != 1))
+    if bad.size:
+        # the reference would silently mis-group any other value (its
+        # label==0 / label==1 masks both miss it); fail loud instead
+        raise ValueError(
+            f"clinical labels must be 0 (good) or 1 (poor): sample "
+            f"{samples[bad[0]]!r} has label {int(labels[bad[0]])}"
+            + (f" (+{bad.size - 1} more)" if bad.size > 1 else ""))
+    return labels
 
 
 def find_common_genes(network_genes, expr_genes) -> List[str]:

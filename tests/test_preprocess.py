@@ -34,3 +34,13 @@ def test_restrict_network_and_data():
 
     eidx = pp.edges_to_indices(rn["edge"], common)
     assert eidx.tolist() == [[0, 1], [2, 0]]
+
+
+def test_match_labels_rejects_non_binary():
+    import numpy as np
+    import pytest
+
+    from g2vec_amd.preprocess import match_labels
+    clinical = {"A": 0, "B": 1, "C": 2}
+    with pytest.raises(ValueError, match="label 2"):
+        match_labels(clinical, np.array(["A", "B", "C"]))
