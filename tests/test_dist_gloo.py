@@ -332,3 +332,46 @@ def test_dp2_minibatch_uneven_shards_no_deadlock():
         p.join(timeout=120)
         assert p.exitcode == 0
     assert 0.0 <= acc <= 1.0
+
+
+def _relabel_worker(rank, world, port, out):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        ps = _pathset(G=50, P=200, seed=17)
+        cfg = G2VecConfig(hidden=64, epochs=8, early_stop=False, seed=4,
+                          device="cpu", dtype="fp32", gene_relabel="on")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                         log=lambda *a, **k: None)
+        res = tr.train(ps)
+        if rank == 0:
+            out.put((res.W_ih.numpy(), res.acc_val_history))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_gene_relabel_matches_single_process():
+    """gene_relabel under DP: rank 0's first-touch order is broadcast so
+    the c all-reduce lives in one id space — results must match the
+    single-process relabeled run (and hence the unrelabeled one)."""
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    out = ctxm.Queue()
+    procs = [ctxm.Process(target=_relabel_worker, args=(r, 2, port, out))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    W_dp, hist_dp = out.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    ps = _pathset(G=50, P=200, seed=17)
+    cfg = G2VecConfig(hidden=64, epochs=8, early_stop=False, seed=4,
+                      device="cpu", dtype="fp32", gene_relabel="on")
+    tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                     log=lambda *a, **k: None)
+    res = tr.train(ps)
+    assert hist_dp == pytest.approx(res.acc_val_history, abs=1e-6)
+    assert np.allclose(W_dp, res.W_ih.numpy(), atol=1e-5)
