@@ -325,11 +325,12 @@ class CbowTrainer:
         c = ops.scatter_dO(tr.genes, tr.offsets, st.dO_buf, self.G,
                            plan=st.plan)
         self.ctx.allreduce_(c)                  # C1: whole dW_ih message
-        ops.gemv_cols(st.W, c, st.gradwho_buf)  # dW_ho = W_ih^T c (pre-update W)
-        ops.adam_rank1(st.W, st.mW, st.vW, c, st.who, st.t_adam, cfg.lr,
-                       self.B1, self.B2, self.EPS, lrt_buf=lrt)
-        ops.adam_dense(st.who, st.mO, st.vO, st.gradwho_buf, st.t_adam,
-                       cfg.lr, self.B1, self.B2, self.EPS, lrt_buf=lrt)
+        # fused tail: one streaming pass over the pre-update W rows does
+        # the rank-1 Adam AND emits dW_ho = W_pre^T c; a fold launch
+        # applies the who update (3 launches + a full W read saved)
+        ops.adam_rank1_fused(st.W, st.mW, st.vW, c, st.who, st.mO, st.vO,
+                             st.t_adam, cfg.lr, self.B1, self.B2, self.EPS,
+                             lrt_buf=lrt)
         # post-update accuracy (reference order, G2Vec.py:264-267): one
         # fused eval kernel over the concatenated train+val paths, which
         # ALSO emits the next epoch's train dlogits (post-update s is the
@@ -935,11 +936,8 @@ class CbowTrainer:
         use_plan = st.plan if (lo == 0 and hi == tr.n_paths) else None
         c = ops.scatter_dO(genes, offsets, dO, self.G, plan=use_plan)
         self.ctx.allreduce_(c)                      # C1: the whole dW_ih message
-        ops.gemv_cols(W, c, st.gradwho_buf)         # dW_ho = W_ih^T c (pre-update W)
-        ops.adam_rank1(W, st.mW, st.vW, c, who, t, self.cfg.lr, self.B1,
-                       self.B2, self.EPS)
-        ops.adam_dense(who, st.mO, st.vO, st.gradwho_buf, t, self.cfg.lr,
-                       self.B1, self.B2, self.EPS)
+        ops.adam_rank1_fused(W, st.mW, st.vW, c, who, st.mO, st.vO, t,
+                             self.cfg.lr, self.B1, self.B2, self.EPS)
 
     def _step_general(self, st, lo, hi, inv_b, t):
         W, W16, who, tr = st.W, st.W16, st.who, st.tr

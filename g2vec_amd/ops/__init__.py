@@ -189,6 +189,26 @@ def adam_rank1(W, m, v, c, who, t: int, lr: float, b1: float, b2: float,
     cpu_ref.adam_rank1(W, m, v, c, who, t, lr, b1, b2, eps)
 
 
+def adam_rank1_fused(W, m, v, c, who, mO, vO, t: int, lr: float,
+                     b1: float, b2: float, eps: float,
+                     lrt_buf: Optional[torch.Tensor] = None) -> None:
+    """Fused epoch tail: rank-1 Adam on W_ih AND the dW_ho = W_pre^T c
+    reduction AND the W_ho Adam update — one streaming pass over the
+    pre-update W rows plus a single fold launch (replaces gemv_cols +
+    adam_rank1 + adam_dense: three launches and an extra full W read).
+    Deterministic (fixed per-thread/block fold order)."""
+    if W.is_cuda:
+        if lrt_buf is None:
+            lrt_buf = torch.tensor([tf1_lr_t(lr, b1, b2, t)],
+                                   dtype=torch.float32, device=W.device)
+        native().adam_rank1(W, m, v, c, who, lrt_buf, float(b1), float(b2),
+                            float(eps), mO=mO, vO=vO)
+        return
+    grad_who = torch.mv(W.t(), c)          # pre-update W
+    cpu_ref.adam_rank1(W, m, v, c, who, t, lr, b1, b2, eps)
+    cpu_ref.adam_dense(who, mO, vO, grad_who, t, lr, b1, b2, eps)
+
+
 def adam_dense(W, m, v, grad, t: int, lr: float, b1: float, b2: float,
                eps: float, lrt_buf: Optional[torch.Tensor] = None) -> None:
     if W.is_cuda:

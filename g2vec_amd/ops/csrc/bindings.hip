@@ -227,7 +227,14 @@ torch::Tensor scatter_dO_det(torch::Tensor inst_path, torch::Tensor seg_start,
 
 void adam_rank1(torch::Tensor W, torch::Tensor m, torch::Tensor v,
                 torch::Tensor c, torch::Tensor who, torch::Tensor lrt,
-                double b1, double b2, double eps) {
+                double b1, double b2, double eps,
+                std::optional<torch::Tensor> mO,
+                std::optional<torch::Tensor> vO) {
+  // mO/vO given: FUSED epoch tail — the same pass that streams the
+  // pre-update W rows for the rank-1 Adam also emits per-block partials
+  // of dW_ho = W_pre^T c, and one fold launch applies the who update
+  // (replaces the separate gemv_cols + fold_cols + adam_dense chain and
+  // its extra full W read; deterministic block order).
   CHECK_DEV(W); CHECK_CONT(W); CHECK_F32(W);
   CHECK_CONT(m); CHECK_CONT(v); CHECK_CONT(c); CHECK_CONT(who);
   CHECK_DEV(lrt); CHECK_F32(lrt);
@@ -236,11 +243,30 @@ void adam_rank1(torch::Tensor W, torch::Tensor m, torch::Tensor v,
   TORCH_CHECK(h % 4 == 0 && 256 % (h / 4) == 0 && h <= 1024,
               "hidden must be a multiple of 4 with h/4 dividing 256");
   const int rpb = 256 / (h / 4);   // rows per 256-thread block
-  hipLaunchKernelGGL(adam_rank1_kernel, dim3(grid_for(G, rpb)), dim3(256), 0,
+  const bool fused = mO.has_value();
+  int grid = grid_for(G, rpb);
+  torch::Tensor partials;
+  float* gw = nullptr;
+  if (fused) {
+    CHECK_DEV(*mO); CHECK_CONT(*mO); CHECK_F32(*mO);
+    CHECK_DEV(*vO); CHECK_CONT(*vO); CHECK_F32(*vO);
+    if (grid > 1024) grid = 1024;    // bound the partial table / fold cost
+    partials = torch::empty({grid, h},
+        torch::TensorOptions().dtype(at::kFloat).device(W.device()));
+    gw = partials.data_ptr<float>();
+  }
+  hipLaunchKernelGGL(adam_rank1_kernel, dim3(grid), dim3(256), 0,
                      cur_stream(), W.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), c.data_ptr<float>(),
                      who.data_ptr<float>(), G, h, lrt.data_ptr<float>(),
-                     (float)b1, (float)b2, (float)eps);
+                     (float)b1, (float)b2, (float)eps, gw);
+  if (fused) {
+    hipLaunchKernelGGL(fold_gw_adam_kernel, dim3(1), dim3(256), 0,
+                       cur_stream(), gw, grid, h, who.data_ptr<float>(),
+                       mO->data_ptr<float>(), vO->data_ptr<float>(),
+                       lrt.data_ptr<float>(), (float)b1, (float)b2,
+                       (float)eps);
+  }
   LAUNCH_CHECK();
 }
 
@@ -587,7 +613,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("p_split"), py::arg("counts"),
         py::arg("dO") = py::none(), py::arg("inv_b") = 1.0,
         "fused train/val correct-count eval (in-place counts[2])");
-  m.def("adam_rank1", &adam_rank1, "TF1 Adam, rank-1 grad");
+  m.def("adam_rank1", &adam_rank1,
+        py::arg("W"), py::arg("m"), py::arg("v"), py::arg("c"),
+        py::arg("who"), py::arg("lrt"), py::arg("b1"), py::arg("b2"),
+        py::arg("eps"), py::arg("mO") = py::none(), py::arg("vO") = py::none(),
+        "TF1 Adam, rank-1 grad (mO/vO given: fused dW_ho + who update)");
   m.def("adam_dense", &adam_dense, "TF1 Adam, dense grad");
   m.def("cbow_fwd", &cbow_fwd,
         py::arg("W"), py::arg("who"), py::arg("genes"), py::arg("offs"),

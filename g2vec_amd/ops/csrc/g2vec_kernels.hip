@@ -492,7 +492,14 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
                   float* __restrict__ v, const float* __restrict__ c,
                   const float* __restrict__ who, long long G, int h,
                   const float* __restrict__ lr_t_ptr, float b1, float b2,
-                  float eps) {
+                  float eps, float* __restrict__ gw_partials) {
+  // gw_partials != nullptr: ALSO emit this block's partial of
+  // dW_ho = W_pre^T c ([gridDim.x, h] layout) — the fused epilogue reads
+  // the pre-update W rows this kernel loads anyway, removing the
+  // separate gemv_cols pass (a full W read) and its fold launch.
+  // Deterministic: each thread accumulates its own grid-stride rows in
+  // order, then fixed ascending-lrow LDS tree per block, then the fold
+  // kernel sums blocks ascending.
   const float lr_t = lr_t_ptr[0];   // device-read so hipGraph replays see
                                     // the per-step bias-corrected value
   f32x4* W4 = (f32x4*)W; f32x4* m4 = (f32x4*)m; f32x4* v4 = (f32x4*)v;
@@ -505,6 +512,7 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
   const int lrow = (int)threadIdx.x / h4;      // this thread's row-in-block
   const int j4 = (int)threadIdx.x - lrow * h4; // and column group
   const f32x4 wj = who4[j4];
+  f32x4 gw_acc = {0.f, 0.f, 0.f, 0.f};
   for (long long g = (long long)blockIdx.x * rpb + lrow; g < G;
        g += (long long)gridDim.x * rpb) {
     const long long i = g * h4 + j4;
@@ -516,6 +524,7 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
     f32x4 ww = __builtin_nontemporal_load(&W4[i]);
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
+      if (gw_partials) gw_acc[k] += cg * ww[k];   // pre-update W row
       const float grad = cg * wj[k];
       mm[k] = b1 * mm[k] + (1.f - b1) * grad;
       vv[k] = b2 * vv[k] + (1.f - b2) * grad * grad;
@@ -524,6 +533,45 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
     __builtin_nontemporal_store(mm, &m4[i]);
     __builtin_nontemporal_store(vv, &v4[i]);
     __builtin_nontemporal_store(ww, &W4[i]);
+  }
+  if (gw_partials) {
+    // per-block fold: slots [rpb][h] in LDS, summed ascending lrow
+    __shared__ float gw_slots[256 * 4];          // rpb * h = 1024 floats
+    float* slot = gw_slots + (lrow * h4 + j4) * 4;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) slot[k] = gw_acc[k];
+    __syncthreads();
+    if (lrow == 0) {
+      f32x4 tot = {0.f, 0.f, 0.f, 0.f};
+      for (int r = 0; r < rpb; ++r) {
+        const float* sl = gw_slots + (r * h4 + j4) * 4;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) tot[k] += sl[k];
+      }
+      float* out = gw_partials + (long long)blockIdx.x * h + j4 * 4;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) out[k] = tot[k];
+    }
+  }
+}
+
+// fold the adam_rank1 gw partials (ascending block order: deterministic)
+// and apply the TF1-Adam update to W_ho in the same launch
+extern "C" __global__ void __launch_bounds__(256)
+fold_gw_adam_kernel(const float* __restrict__ partials, int n_blocks, int h,
+                    float* __restrict__ who, float* __restrict__ mO,
+                    float* __restrict__ vO,
+                    const float* __restrict__ lr_t_ptr, float b1, float b2,
+                    float eps) {
+  const float lr_t = lr_t_ptr[0];
+  for (int i = threadIdx.x; i < h; i += blockDim.x) {
+    float g = 0.f;
+    for (int b = 0; b < n_blocks; ++b) g += partials[(long long)b * h + i];
+    float mm = b1 * mO[i] + (1.f - b1) * g;
+    float vv = b2 * vO[i] + (1.f - b2) * g * g;
+    mO[i] = mm;
+    vO[i] = vv;
+    who[i] -= lr_t * mm / (sqrtf(vv) + eps);
   }
 }
 

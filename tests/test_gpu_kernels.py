@@ -366,3 +366,46 @@ def test_relu_general_kernels_match_oracle_on_gpu():
     assert torch.equal(corr_g.cpu(), corr_c)
     assert torch.allclose(H_g.cpu(), H_c, atol=1e-5)
     assert torch.allclose(dW_g.cpu(), dW_c, atol=1e-5)
+
+
+@pytest.mark.timeout(600)
+def test_adam_rank1_fused_matches_unfused_sequence():
+    """The fused epoch tail (rank-1 Adam + in-pass dW_ho partials + who
+    fold/update) vs the explicit gemv_cols + adam_rank1 + adam_dense
+    sequence: W/m/v bitwise (same math), who/mO/vO within fp32
+    reduction-order tolerance (the dW_ho summation tree differs)."""
+    dev = torch.device("cuda")
+    for h in (64, 128, 512):
+        G = 5000
+        gen = torch.Generator(device="cpu").manual_seed(h)
+        W0 = torch.randn(G, h, generator=gen).to(dev) * 0.1
+        m0 = torch.randn(G, h, generator=gen).to(dev) * 0.01
+        v0 = torch.rand(G, h, generator=gen).to(dev) * 0.01
+        c = torch.randn(G, generator=gen).to(dev) * 0.1
+        who0 = torch.randn(h, generator=gen).to(dev) * 0.1
+        mO0 = torch.randn(h, generator=gen).to(dev) * 0.01
+        vO0 = torch.rand(h, generator=gen).to(dev) * 0.01
+        lrt = torch.tensor([0.003], device=dev)
+
+        Wa, ma, va = W0.clone(), m0.clone(), v0.clone()
+        whoa, mOa, vOa = who0.clone(), mO0.clone(), vO0.clone()
+        gw = torch.empty(h, device=dev)
+        ops.native().gemv_cols_(Wa, c, gw)
+        ops.native().adam_rank1(Wa, ma, va, c, whoa, lrt, 0.9, 0.999, 1e-8)
+        ops.native().adam_dense(whoa, mOa, vOa, gw, lrt, 0.9, 0.999, 1e-8)
+
+        Wb, mb, vb = W0.clone(), m0.clone(), v0.clone()
+        whob, mOb, vOb = who0.clone(), mO0.clone(), vO0.clone()
+        ops.native().adam_rank1(Wb, mb, vb, c, whob, lrt, 0.9, 0.999, 1e-8,
+                                mO=mOb, vO=vOb)
+        assert torch.equal(Wa, Wb) and torch.equal(ma, mb)
+        assert torch.equal(va, vb)
+        assert torch.allclose(whoa, whob, atol=1e-6)
+        assert torch.allclose(mOa, mOb, atol=1e-6)
+        assert torch.allclose(vOa, vOb, atol=1e-6)
+        # determinism: repeat the fused call from the same state
+        Wc, mc, vc = W0.clone(), m0.clone(), v0.clone()
+        whoc, mOc, vOc = who0.clone(), mO0.clone(), vO0.clone()
+        ops.native().adam_rank1(Wc, mc, vc, c, whoc, lrt, 0.9, 0.999, 1e-8,
+                                mO=mOc, vO=vOc)
+        assert torch.equal(whob, whoc) and torch.equal(mOb, mOc)
