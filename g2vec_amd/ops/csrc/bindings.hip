@@ -261,8 +261,8 @@ void adam_rank1(torch::Tensor W, torch::Tensor m, torch::Tensor v,
                      who.data_ptr<float>(), G, h, lrt.data_ptr<float>(),
                      (float)b1, (float)b2, (float)eps, gw);
   if (fused) {
-    hipLaunchKernelGGL(fold_gw_adam_kernel, dim3(1), dim3(256), 0,
-                       cur_stream(), gw, grid, h, who.data_ptr<float>(),
+    hipLaunchKernelGGL(fold_gw_adam_kernel, dim3(grid_for(h, 4)), dim3(256),
+                       0, cur_stream(), gw, grid, h, who.data_ptr<float>(),
                        mO->data_ptr<float>(), vO->data_ptr<float>(),
                        lrt.data_ptr<float>(), (float)b1, (float)b2,
                        (float)eps);

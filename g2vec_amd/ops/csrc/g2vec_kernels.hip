@@ -555,8 +555,10 @@ adam_rank1_kernel(float* __restrict__ W, float* __restrict__ m,
   }
 }
 
-// fold the adam_rank1 gw partials (ascending block order: deterministic)
-// and apply the TF1-Adam update to W_ho in the same launch
+// fold the adam_rank1 gw partials and apply the TF1-Adam update to W_ho
+// in the same launch. One WAVE per column: lanes stride the block
+// partials (fixed stride + fixed shuffle tree -> deterministic); the
+// single-block variant was latency-bound at ~1k dependent loads/column.
 extern "C" __global__ void __launch_bounds__(256)
 fold_gw_adam_kernel(const float* __restrict__ partials, int n_blocks, int h,
                     float* __restrict__ who, float* __restrict__ mO,
@@ -564,14 +566,21 @@ fold_gw_adam_kernel(const float* __restrict__ partials, int n_blocks, int h,
                     const float* __restrict__ lr_t_ptr, float b1, float b2,
                     float eps) {
   const float lr_t = lr_t_ptr[0];
-  for (int i = threadIdx.x; i < h; i += blockDim.x) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  for (int i = blockIdx.x * wpb + wib; i < h; i += gridDim.x * wpb) {
     float g = 0.f;
-    for (int b = 0; b < n_blocks; ++b) g += partials[(long long)b * h + i];
-    float mm = b1 * mO[i] + (1.f - b1) * g;
-    float vv = b2 * vO[i] + (1.f - b2) * g * g;
-    mO[i] = mm;
-    vO[i] = vv;
-    who[i] -= lr_t * mm / (sqrtf(vv) + eps);
+    for (int b = lane; b < n_blocks; b += WAVE)
+      g += partials[(long long)b * h + i];
+    g = wave_sum(g);
+    if (lane == 0) {
+      const float mm = b1 * mO[i] + (1.f - b1) * g;
+      const float vv = b2 * vO[i] + (1.f - b2) * g * g;
+      mO[i] = mm;
+      vO[i] = vv;
+      who[i] -= lr_t * mm / (sqrtf(vv) + eps);
+    }
   }
 }
 
