@@ -315,7 +315,12 @@ class CbowTrainer:
         tr, vl = st.tr, st.vl
         if counts_out is None:
             counts_out = st.counts_buf
-        counts_out.zero_()
+        if self.device.type != "cuda" or st.ev_genes is None:
+            counts_out.zero_()      # CPU oracle accumulates (+=); the GPU
+                                    # fold kernel OVERWRITES counts, so the
+                                    # fill launch is pure overhead there
+                                    # (still zeroed on an empty-shard rank,
+                                    # whose eval never runs)
         if lrt_slot is not None:
             lrt = lrt_slot
         else:
