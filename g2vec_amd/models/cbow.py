@@ -254,7 +254,6 @@ class CbowTrainer:
         st.epoch_idx = 0
         # persistent fast-path buffers (stable addresses across hipGraph replays)
         st.s_buf = None
-        st.gradwho_buf = torch.empty_like(who)
         if not use_general:
             st.s_buf = torch.empty(self.G, dtype=torch.float32,
                                    device=self.device)
