@@ -36,12 +36,10 @@ import numpy as np
 import torch
 
 from g2vec_amd.config import G2VecConfig
-from g2vec_amd.graph import build_group_graph
 from g2vec_amd.models.cbow import CbowTrainer
-from g2vec_amd.parallel.dist import init_dist
-from g2vec_amd.paths import integrate_pathsets
+from g2vec_amd.parallel.dist import init_dist, single
+from g2vec_amd.pipeline import generate_paths
 from g2vec_amd.utils import synth
-from g2vec_amd.walks import generate_walks
 
 BASELINE_PATHS_PER_SEC = 16500.0   # BASELINE.md derived reference number
 
@@ -156,33 +154,31 @@ def main() -> int:
     labels_t = torch.from_numpy(labels).to(device)
     edge_t = torch.from_numpy(edge_idx).to(device)
 
+    # weak scaling: each rank walks its OWN dataset with full source
+    # coverage — a single-process context (no C5 sharding/gather here)
+    sctx = single(device)
     if on_gpu:
         # one-time hipModule/dispatcher loads + allocator arenas for the
         # step-3 op set (~1.2 s on a fresh process,
         # profiles/step3_cold.json) happen in a full-size dummy round so
         # walks_per_sec reports the pipeline, not runtime init
-        ws_warm = []
-        for group in (0, 1):
-            gw = build_group_graph(expr_t, labels_t, group, edge_t, n_genes)
-            ws_warm.append(generate_walks(gw, cfg.len_path,
-                                          cfg.num_repetition, 999, group))
-        integrate_pathsets(ws_warm[0], ws_warm[1], n_genes)
-        del ws_warm
+        warm_cfg = dataclasses.replace(cfg, seed=999)
+        generate_paths(warm_cfg, expr_t, labels_t, edge_t, n_genes, sctx,
+                       log=(lambda *a, **k2: None))
         torch.cuda.synchronize()
 
     t0 = time.perf_counter()
-    walksets = []
-    for group in (0, 1):
-        g = build_group_graph(expr_t, labels_t, group, edge_t, n_genes)
-        walksets.append(generate_walks(g, cfg.len_path, cfg.num_repetition,
-                                       cfg.seed + rank, group))
+    ps, _freq, _nip, stats = generate_paths(
+        cfg, expr_t, labels_t, edge_t, n_genes, sctx,
+        log=(lambda *a, **k2: None))
     if on_gpu:
         torch.cuda.synchronize()
-    walk_s = time.perf_counter() - t0
-    n_walks = sum(int(w.nodes.shape[0]) for w in walksets)
-    ps, _freq, _nip = integrate_pathsets(walksets[0], walksets[1], n_genes)
+    walk_s = time.perf_counter() - t0   # END-TO-END step 3: both groups'
+                                        # graph builds + walks + dedup
+    n_walks = int(stats["n_walks"])
     log(f"[bench] rank {rank}: {n_walks} walks in {walk_s:.3f}s "
-        f"({n_walks / walk_s:.0f} walks/s), {ps.n_paths} paths after dedup")
+        f"({n_walks / walk_s:.0f} walks/s end-to-end), {ps.n_paths} paths "
+        f"after dedup")
 
     # ---- convergence probe: epochs + wall to val-ACC >= 0.88, over
     # --conv-seeds independent training seeds (same dataset; fresh split

@@ -44,17 +44,47 @@ def generate_paths(cfg: G2VecConfig, expr_t: torch.Tensor, labels_t: torch.Tenso
     seed = cfg.seed if cfg.seed is not None else int(time.time_ns() & 0x7FFFFFFF)
     walksets = []
     stats: Dict[str, float] = {}
-    for group in (0, 1):
-        with timers.phase(f"graph_g{group}"):
-            g = build_group_graph(expr_t, labels_t, group, edge_idx_t, n_genes,
-                                  threshold=cfg.pcc_threshold, mode=cfg.pcc_mode)
-        lo, hi = ctx.shard_range(n_genes)
-        with timers.phase(f"walks_g{group}"):
-            ws = generate_walks(g, cfg.len_path, cfg.num_repetition, seed,
-                                group, (lo, hi))
-            ws = _gather_walks(ctx, ws)
-        walksets.append(ws)
-        stats[f"nnz_g{group}"] = int(g.col_idx.numel())
+    lo, hi = ctx.shard_range(n_genes)
+    if expr_t.is_cuda:
+        # the two prognosis groups' graph-build + walk chains are
+        # independent: run them on separate HIP streams so each group's
+        # PCC/threshold/CSR torch chains and walk kernel overlap the
+        # other's (the C5 all-gather stays on the default stream, after
+        # both streams join). Bitwise identical output — only scheduling
+        # changes.
+        streams = [torch.cuda.Stream(), torch.cuda.Stream()]
+        graphs = [None, None]
+        raw = [None, None]
+        with timers.phase("graphs_walks_overlapped"):
+            for group in (0, 1):
+                streams[group].wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(streams[group]):
+                    g = build_group_graph(expr_t, labels_t, group,
+                                          edge_idx_t, n_genes,
+                                          threshold=cfg.pcc_threshold,
+                                          mode=cfg.pcc_mode)
+                    graphs[group] = g
+                    raw[group] = generate_walks(g, cfg.len_path,
+                                                cfg.num_repetition, seed,
+                                                group, (lo, hi))
+            for st_ in streams:
+                torch.cuda.current_stream().wait_stream(st_)
+        for group in (0, 1):
+            with timers.phase(f"walks_g{group}"):
+                walksets.append(_gather_walks(ctx, raw[group]))
+            stats[f"nnz_g{group}"] = int(graphs[group].col_idx.numel())
+    else:
+        for group in (0, 1):
+            with timers.phase(f"graph_g{group}"):
+                g = build_group_graph(expr_t, labels_t, group, edge_idx_t,
+                                      n_genes, threshold=cfg.pcc_threshold,
+                                      mode=cfg.pcc_mode)
+            with timers.phase(f"walks_g{group}"):
+                ws = generate_walks(g, cfg.len_path, cfg.num_repetition, seed,
+                                    group, (lo, hi))
+                ws = _gather_walks(ctx, ws)
+            walksets.append(ws)
+            stats[f"nnz_g{group}"] = int(g.col_idx.numel())
     with timers.phase("integrate"):
         ps, freq, n_in_paths = integrate_pathsets(walksets[0], walksets[1],
                                                   n_genes)
