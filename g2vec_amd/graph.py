@@ -58,22 +58,32 @@ def edge_pcc_weights(zt: torch.Tensor, edge_idx: torch.Tensor,
     return ops.pcc_edges(zt, edge_idx, n_group)
 
 
+def dedupe_edges(edge_idx: torch.Tensor, n_genes: int) -> torch.Tensor:
+    """Unique directed edges as i32 [E,2] (the dense adjMat overwrites
+    repeated file edges, G2Vec.py:390). Dataset-static: compute once and
+    pass to build_group_graph for both prognosis groups."""
+    key = edge_idx[:, 0].long() * n_genes + edge_idx[:, 1].long()
+    key = torch.unique(key)
+    src = (key // n_genes).int()
+    dst = (key % n_genes).int()
+    return torch.stack([src, dst], dim=1)
+
+
 def build_group_graph(expr: torch.Tensor, labels: torch.Tensor, group: int,
                       edge_idx: torch.Tensor, n_genes: int,
-                      threshold: float = 0.5, mode: str = "auto") -> CsrGraph:
+                      threshold: float = 0.5, mode: str = "auto",
+                      edges_deduped: bool = False) -> CsrGraph:
     """CSR graph for one prognosis group (the sparse equivalent of
-    construct_adjMat, G2Vec.py:370-391)."""
+    construct_adjMat, G2Vec.py:370-391). edges_deduped=True skips the
+    per-call unique pass (callers hoist dedupe_edges once per dataset)."""
     device = expr.device
     if edge_idx.numel() == 0:
         return CsrGraph(torch.zeros(n_genes + 1, dtype=torch.int32, device=device),
                         torch.zeros(0, dtype=torch.int32, device=device),
                         torch.zeros(0, dtype=torch.float32, device=device), n_genes)
-    # deduplicate repeated file edges (dense adjMat overwrites the same cell)
-    key = edge_idx[:, 0].long() * n_genes + edge_idx[:, 1].long()
-    key = torch.unique(key)
-    src = (key // n_genes).int()
-    dst = (key % n_genes).int()
-    pairs = torch.stack([src, dst], dim=1)
+    pairs = (edge_idx.int() if edges_deduped
+             else dedupe_edges(edge_idx, n_genes))
+    src, dst = pairs[:, 0].contiguous(), pairs[:, 1].contiguous()
 
     zt = zscore_group(expr, labels, group)
     w = edge_pcc_weights(zt, pairs, mode)

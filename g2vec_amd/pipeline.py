@@ -45,6 +45,8 @@ def generate_paths(cfg: G2VecConfig, expr_t: torch.Tensor, labels_t: torch.Tenso
     walksets = []
     stats: Dict[str, float] = {}
     lo, hi = ctx.shard_range(n_genes)
+    from .graph import dedupe_edges
+    edge_idx_t = dedupe_edges(edge_idx_t, n_genes)   # once for both groups
     if expr_t.is_cuda:
         # the two prognosis groups' graph-build + walk chains are
         # independent: run them on separate HIP streams so each group's
@@ -62,7 +64,8 @@ def generate_paths(cfg: G2VecConfig, expr_t: torch.Tensor, labels_t: torch.Tenso
                     g = build_group_graph(expr_t, labels_t, group,
                                           edge_idx_t, n_genes,
                                           threshold=cfg.pcc_threshold,
-                                          mode=cfg.pcc_mode)
+                                          mode=cfg.pcc_mode,
+                                          edges_deduped=True)
                     graphs[group] = g
                     raw[group] = generate_walks(g, cfg.len_path,
                                                 cfg.num_repetition, seed,
@@ -78,7 +81,7 @@ def generate_paths(cfg: G2VecConfig, expr_t: torch.Tensor, labels_t: torch.Tenso
             with timers.phase(f"graph_g{group}"):
                 g = build_group_graph(expr_t, labels_t, group, edge_idx_t,
                                       n_genes, threshold=cfg.pcc_threshold,
-                                      mode=cfg.pcc_mode)
+                                      mode=cfg.pcc_mode, edges_deduped=True)
             with timers.phase(f"walks_g{group}"):
                 ws = generate_walks(g, cfg.len_path, cfg.num_repetition, seed,
                                     group, (lo, hi))
