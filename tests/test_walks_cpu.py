@@ -79,3 +79,26 @@ def test_generate_walks_sharding():
     lo = generate_walks(g, 4, 2, seed=1, group=0, src_range=(0, 2))
     assert lo.nodes.shape[0] == 4
     assert set(lo.nodes[:, 0].tolist()) == {0, 1}
+
+
+def test_self_loops_never_revisited():
+    """A self-loop edge (absent from the reference data but legal in user
+    files) is a visited candidate from step one — the non-revisit mask
+    must keep the walker off it."""
+    import torch
+
+    from g2vec_amd import ops
+    G = 6
+    # each node: self-loop + edge to next
+    row_ptr = torch.arange(0, 2 * G + 1, 2, dtype=torch.int32)
+    col, w = [], []
+    for i in range(G):
+        col += [i, (i + 1) % G]
+        w += [5.0, 1.0]
+    nodes, lengths, _ = ops.random_walks(
+        torch.tensor(row_ptr.tolist(), dtype=torch.int32),
+        torch.tensor(col, dtype=torch.int32),
+        torch.tensor(w), torch.arange(G, dtype=torch.int32), 3, 8, seed=5)
+    for k in range(nodes.shape[0]):
+        path = nodes[k, :int(lengths[k])].tolist()
+        assert len(path) == len(set(path))
