@@ -300,6 +300,7 @@ def main() -> int:
                                           if cross_ep else None),
                 "conv_runs": conv_runs,
                 "walks_per_sec": round(n_walks / walk_s, 1),
+                "grad_allreduce_bytes_per_epoch": n_genes * 4,
                 "step_includes": "full-batch fwd+bwd+allreduce+dense-Adam "
                                  "+ post-update train/val ACC evals",
             },
