@@ -375,3 +375,53 @@ def test_dp2_gene_relabel_matches_single_process():
     res = tr.train(ps)
     assert hist_dp == pytest.approx(res.acc_val_history, abs=1e-6)
     assert np.allclose(W_dp, res.W_ih.numpy(), atol=1e-5)
+
+
+def _general_worker(rank, world, port, out):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        ps = _pathset(G=40, P=150, seed=23)
+        for act in ("none", "relu"):
+            cfg = G2VecConfig(hidden=64, epochs=5, early_stop=False, seed=4,
+                              device="cpu", trainer_path="general",
+                              activation=act)
+            tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                             log=lambda *a, **k: None)
+            res = tr.train(ps)
+            if rank == 0:
+                out.put((act, res.W_ih.numpy(), res.acc_val_history))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_general_path_matches_single_process():
+    """The general kernel chain (dense dW + grad_who all-reduces) under
+    world_size=2, linear AND relu — must match single-process."""
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    out = ctxm.Queue()
+    procs = [ctxm.Process(target=_general_worker, args=(r, 2, port, out))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(2):
+        act, W, hist = out.get(timeout=240)
+        got[act] = (W, hist)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    ps = _pathset(G=40, P=150, seed=23)
+    for act in ("none", "relu"):
+        cfg = G2VecConfig(hidden=64, epochs=5, early_stop=False, seed=4,
+                          device="cpu", trainer_path="general",
+                          activation=act)
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"),
+                         log=lambda *a, **k: None)
+        res = tr.train(ps)
+        W_dp, hist_dp = got[act]
+        assert hist_dp == pytest.approx(res.acc_val_history, abs=1e-6), act
+        assert np.allclose(W_dp, res.W_ih.numpy(), atol=1e-5), act
