@@ -273,3 +273,29 @@ def test_failure_recovery_paths_cache_plus_train_resume(tiny_files, tmp_path):
            (tmp_path / "full_vectors.txt").read_text()
     assert (tmp_path / "rec_biomarkers.txt").read_text() == \
            (tmp_path / "full_biomarkers.txt").read_text()
+
+
+def test_compare_outputs_tool(tiny_files, tmp_path):
+    """tools/compare_outputs.py: exit 0 on identical triples, 1 on a
+    perturbed one, with a per-file report."""
+    import subprocess
+    import sys as _sys
+
+    run(_cfg(tiny_files, tmp_path, result_name=str(tmp_path / "a"), epochs=4))
+    run(_cfg(tiny_files, tmp_path, result_name=str(tmp_path / "b"), epochs=4))
+    r = subprocess.run([_sys.executable, "tools/compare_outputs.py",
+                        str(tmp_path / "a"), str(tmp_path / "b")],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "MATCH" in r.stdout
+    # perturb one lgroup assignment
+    p = tmp_path / "b_lgroups.txt"
+    lines = p.read_text().splitlines()
+    g, v = lines[1].split("\t")
+    lines[1] = f"{g}\t{(int(v) + 1) % 3}"
+    p.write_text("\n".join(lines) + "\n")
+    r = subprocess.run([_sys.executable, "tools/compare_outputs.py",
+                        str(tmp_path / "a"), str(tmp_path / "b")],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 1
+    assert "DIFFER" in r.stdout
