@@ -244,3 +244,27 @@ def test_train_state_resume_bitwise_on_gpu(tmp_path):
     assert resumed.acc_val_history[8:] == pytest.approx(
         full.acc_val_history[8:], abs=0)
     assert torch.equal(resumed.W_ih.cpu(), full.W_ih.cpu())
+
+
+@pytest.mark.timeout(600)
+def test_cli_end_to_end_on_gpu(tmp_path):
+    """The user-facing CLI on cuda: `python -m g2vec_amd ...` end to end,
+    reference transcript + output triple."""
+    import subprocess
+    import sys as _sys
+
+    from g2vec_amd.utils.synth import make_ex_style_files
+    files = make_ex_style_files(str(tmp_path), n_genes=800, n_extra=80,
+                                n_edges=20000, n_samples=100, n_poor=43,
+                                n_modules=8, seed=2)
+    r = subprocess.run(
+        [_sys.executable, "-m", "g2vec_amd", files["expression"],
+         files["clinical"], files["network"], str(tmp_path / "out"),
+         "-p", "30", "-r", "3", "-e", "25", "--seed", "0",
+         "--device", "cuda"],
+        capture_output=True, text=True, timeout=500)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "Optimization Finish" in r.stdout
+    assert ">>> 7. Save results" in r.stdout
+    for sfx in ("_biomarkers.txt", "_lgroups.txt", "_vectors.txt"):
+        assert (tmp_path / f"out{sfx}").exists()
