@@ -65,7 +65,11 @@ def build_parser() -> argparse.ArgumentParser:
                    help="read early-stop accuracies every K epochs and "
                         "deterministically replay to the dip on stop "
                         "(same trajectory/stop/weights; 1/K the "
-                        "collectives and readbacks; GPU fast path)")
+                        "collectives and readbacks; fast path)")
+    p.add_argument("--gene-relabel", choices=["auto", "on", "off"],
+                   default="auto",
+                   help="trainer-internal gather-locality gene relabeling "
+                        "(auto = on at >= 100k genes; pure layout change)")
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph capture of the training epoch")
     p.add_argument("--save-paths", type=str, default="")
@@ -89,7 +93,7 @@ def args_to_config(a: argparse.Namespace) -> G2VecConfig:
         seed=(None if a.seed < 0 else a.seed), dtype=a.dtype, device=a.device,
         pcc_mode=a.pcc_mode, pcc_threshold=a.pcc_threshold,
         kmeans_backend=a.kmeans, trainer_path=a.trainer_path,
-        activation=a.activation,
+        activation=a.activation, gene_relabel=a.gene_relabel,
         batch_size=a.batch_size, compat_lgroup_bug=a.compat_lgroup_bug,
         early_stop=not a.no_early_stop, earlystop_every=a.earlystop_every,
         train_ckpt=a.train_ckpt, train_ckpt_every=a.train_ckpt_every,
