@@ -223,7 +223,6 @@ def test_degenerate_inputs_fail_loudly(tiny_files, tmp_path):
     actionable messages instead of nan-propagating (the reference's
     behavior) or crashing obscurely."""
     # (a) one-class clinical file
-    import shutil
     bad_cli = tmp_path / "bad_CLINICAL.txt"
     lines = open(tiny_files["clinical"]).read().splitlines()
     with open(bad_cli, "w") as f:
