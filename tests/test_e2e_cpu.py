@@ -298,3 +298,28 @@ def test_compare_outputs_tool(tiny_files, tmp_path):
                        capture_output=True, text=True, timeout=120)
     assert r.returncode == 1
     assert "DIFFER" in r.stdout
+
+
+def test_golden_output_triple(tmp_path):
+    """Byte-exact regression pin: the seeded tiny pipeline reproduces
+    the committed golden triple (tests/golden/). Guards every layer that
+    shapes outputs — parsers, graph, walks, trainer numerics, L-groups,
+    scoring, writers. If an INTENTIONAL semantic change breaks this,
+    regenerate the goldens with the snippet in tests/golden/README."""
+    from g2vec_amd.utils.synth import make_ex_style_files
+    files = make_ex_style_files(str(tmp_path), n_genes=120, n_extra=20,
+                                n_edges=2500, n_samples=60, n_poor=26,
+                                n_modules=5, seed=9)
+    cfg = G2VecConfig(expression_file=files["expression"],
+                      clinical_file=files["clinical"],
+                      network_file=files["network"],
+                      result_name=str(tmp_path / "g"),
+                      len_path=12, num_repetition=3, hidden=64, epochs=15,
+                      num_biomarker=10, seed=7, device="cpu")
+    run(cfg)
+    import pathlib
+    golden = pathlib.Path(__file__).parent / "golden"
+    for sfx in ("biomarkers", "lgroups", "vectors"):
+        got = (tmp_path / f"g_{sfx}.txt").read_text()
+        want = (golden / f"tiny_{sfx}.txt").read_text()
+        assert got == want, f"{sfx} drifted from the golden pin"
