@@ -296,6 +296,9 @@ def main() -> int:
                 "conv_crossed": len(crossed),
                 "wall_to_0.88_s_median": (round(float(np.median(crossed)), 4)
                                           if crossed else None),
+                "wall_to_0.88_speedup_vs_ref": (
+                    round(56.7 / float(np.median(crossed)), 1)
+                    if crossed else None),   # reference: ~56.7 s
                 "epochs_to_0.88_median": (float(np.median(cross_ep))
                                           if cross_ep else None),
                 "conv_runs": conv_runs,
