@@ -65,12 +65,19 @@ def test_cli_surface(tiny_files, tmp_path):
     argv = [tiny_files["expression"], tiny_files["clinical"],
             tiny_files["network"], str(tmp_path / "o"),
             "-p", "20", "-r", "2", "-s", "64", "-e", "10", "-l", "0.01",
-            "-n", "7", "--device", "cpu", "--compat-lgroup-bug"]
+            "-n", "7", "--device", "cpu", "--compat-lgroup-bug",
+            "--gene-relabel", "off", "--trainer-path", "general",
+            "--activation", "relu", "--train-ckpt", "ck.pt",
+            "--train-ckpt-every", "3", "--earlystop-every", "4"]
     cfg = args_to_config(build_parser().parse_args(argv))
     assert cfg.len_path == 20 and cfg.num_repetition == 2
     assert cfg.hidden == 64 and cfg.epochs == 10
     assert cfg.lr == 0.01 and cfg.num_biomarker == 7
     assert cfg.compat_lgroup_bug
+    assert cfg.gene_relabel == "off" and cfg.activation == "relu"
+    assert cfg.train_ckpt == "ck.pt" and cfg.train_ckpt_every == 3
+    assert cfg.earlystop_every == 4
+    cfg.validate()
 
 
 def test_jsonl_metrics(tiny_files, tmp_path):
