@@ -167,7 +167,7 @@ def test_adam_kernels_match_oracle():
     assert torch.allclose(W2g.cpu(), W2, atol=1e-6)
 
 
-@pytest.mark.parametrize("hidden", [64, 128, 256])
+@pytest.mark.parametrize("hidden", [64, 128, 256, 1024])
 @pytest.mark.parametrize("dtype", ["fp32", "bf16", "fp16"])
 def test_cbow_fwd_general_matches_oracle(hidden, dtype):
     genes, offs, labels = _pathset_tensors(seed=4)
@@ -375,7 +375,7 @@ def test_adam_rank1_fused_matches_unfused_sequence():
     sequence: W/m/v bitwise (same math), who/mO/vO within fp32
     reduction-order tolerance (the dW_ho summation tree differs)."""
     dev = torch.device("cuda")
-    for h in (64, 128, 512):
+    for h in (64, 128, 512, 1024):
         G = 5000
         gen = torch.Generator(device="cpu").manual_seed(h)
         W0 = torch.randn(G, h, generator=gen).to(dev) * 0.1
