@@ -17,6 +17,7 @@ setup(
     name="g2vec_amd",
     version="0.1.0",
     packages=find_packages(include=["g2vec_amd", "g2vec_amd.*"]),
+    package_data={"g2vec_amd": ["data/*.npz"]},
     ext_modules=[
         CUDAExtension(
             name="g2vec_amd._C",
