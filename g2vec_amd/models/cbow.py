@@ -169,7 +169,9 @@ class CbowTrainer:
         (profiles/README.md, 1M-config notes). Untouched genes keep
         their relative order at the end. Rank 0's order is broadcast so
         every DP rank maps identically (the c all-reduce lives in the
-        relabeled id space)."""
+        relabeled id space); when weak-scaling ranks hold INDEPENDENT
+        datasets, rank 0's order is correct-but-approximate locality for
+        the other ranks."""
         big = torch.iinfo(torch.int64).max
         first = torch.full((self.G,), big, dtype=torch.int64,
                            device=ps.genes.device)
