@@ -117,6 +117,39 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   TORCH_CHECK(counts.numel() == 2, "counts must have 2 elements");
   const long long P = labels.numel();
   if (P == 0) return;
+  // LDS-staged variant (G2VEC_EVAL_LDS=<bytes> opts in when the whole s
+  // table fits that per-block LDS budget): stages s in LDS so gathers are
+  // bank-parallel ds_reads instead of L1 line-divergent loads, grid
+  // capped (G2VEC_EVAL_LDS_GRID, default 1024) so each block amortizes
+  // its stage. Bitwise-identical math. Measured (profiles/README.md,
+  // round 2): wins the isolated microbench ~11% at ex_* shape (27.2 vs
+  // 30.2 us, tools/bench_eval.py) but is NEUTRAL-to-slightly-worse
+  // inside the real epoch chain (905 -> 893-902M paths/s same-box), so
+  // it is OFF by default — kept as the documented measured alternative.
+  const long long G = s.numel();
+  const char* lds_env = getenv("G2VEC_EVAL_LDS");
+  const long long lds_cap = lds_env ? atoll(lds_env) : 0;
+  if (G * 4 <= lds_cap && G * 4 <= 158 * 1024) {   // 160 KB LDS hard limit
+    const char* ge = getenv("G2VEC_EVAL_LDS_GRID");
+    int grid = grid_for(P, 16);
+    const int gcap = ge ? atoi(ge) : 1024;
+    if (grid > gcap) grid = gcap;
+    auto partials = torch::empty({grid, 2},
+        torch::TensorOptions().dtype(at::kFloat).device(s.device()));
+    hipLaunchKernelGGL(cbow_eval_counts_lds_kernel, dim3(grid),
+                       dim3(256), (size_t)(G * 4), cur_stream(),
+                       s.data_ptr<float>(),
+                       genes.data_ptr<int>(), offs.data_ptr<int>(),
+                       labels.data_ptr<float>(), P, (long long)p_split,
+                       partials.data_ptr<float>(),
+                       dO ? dO->data_ptr<float>() : nullptr, (float)inv_b,
+                       (int)G);
+    hipLaunchKernelGGL(fold_partials_kernel, dim3(1), dim3(256), 0,
+                       cur_stream(), partials.data_ptr<float>(), grid,
+                       counts.data_ptr<float>());
+    LAUNCH_CHECK();
+    return;
+  }
   int grid = grid_for(P, 16);       // 16 paths per block (16-lane sub-waves)
   if (grid > 2048) grid = 2048;      // 8192 waves fill the chip
   auto partials = torch::empty({grid, 2},

@@ -429,6 +429,59 @@ cbow_eval_counts_kernel(const float* __restrict__ s, const int* __restrict__ gen
   }
 }
 
+// LDS-staged variant: at ex_* scale the whole s table is 30 KB, but the
+// subwave gathers above are TA-throughput bound — a 64-lane gather over a
+// 30 KB range touches ~40-60 distinct L1 lines, serialized in the CU's
+// one texture/addressing unit. Staging s in LDS turns each gather into a
+// ds_read (32-bank parallel, ~2-4x conflict factor), so the block pays
+// one coalesced G-float stage (L2-broadcast: every block reads the same
+// table) and then gathers at LDS rate. The per-path math and accumulation
+// order are IDENTICAL to cbow_eval_counts_kernel — bitwise-equal outputs;
+// the host picks this kernel when G*4 fits the LDS budget and caps the
+// grid so each resident block amortizes its stage over many paths.
+extern "C" __global__ void __launch_bounds__(256)
+cbow_eval_counts_lds_kernel(const float* __restrict__ s,
+                            const int* __restrict__ genes,
+                            const int* __restrict__ offs,
+                            const float* __restrict__ labels,
+                            long long P, long long p_split,
+                            float* __restrict__ partials,
+                            float* __restrict__ dO, float inv_b, int G) {
+  extern __shared__ float s_lds[];
+  for (int g = threadIdx.x; g < G; g += blockDim.x) s_lds[g] = s[g];
+  __syncthreads();
+  const int sublane = threadIdx.x & (SUBW - 1);
+  const int subs_per_block = blockDim.x / SUBW;
+  const int sub = threadIdx.x / SUBW;
+  float c0 = 0.f, c1 = 0.f;
+  for (long long p = (long long)blockIdx.x * subs_per_block + sub; p < P;
+       p += (long long)gridDim.x * subs_per_block) {
+    const int lo = offs[p], hi = offs[p + 1];
+    float partial = 0.f;
+    for (int i = lo + sublane; i < hi; i += SUBW)
+      partial += s_lds[__builtin_nontemporal_load(&genes[i])];
+    const float o = subwave_sum16(partial);
+    if (sublane == 0) {
+      const float y = labels[p];
+      const float corr = (((o > 0.f ? 1.f : 0.f) == y) ? 1.f : 0.f);
+      if (p < p_split) c0 += corr; else c1 += corr;
+      if (dO && p < p_split)
+        dO[p] = (1.f / (1.f + expf(-o)) - y) * inv_b;
+    }
+  }
+  c0 = wave_sum(c0);
+  c1 = wave_sum(c1);
+  __shared__ float sm[8];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wib = threadIdx.x >> 6;
+  if (lane == 0) { sm[wib] = c0; sm[4 + wib] = c1; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    partials[2 * blockIdx.x] = sm[0] + sm[1] + sm[2] + sm[3];
+    partials[2 * blockIdx.x + 1] = sm[4] + sm[5] + sm[6] + sm[7];
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 fold_partials_kernel(const float* __restrict__ partials, int n_blocks,
                      float* __restrict__ counts) {

@@ -46,11 +46,26 @@ def main():
         print(f"{name:24s} {(time.perf_counter()-t0)/iters*1e6:9.2f} us "
               f"counts={counts.tolist()}", flush=True)
 
-    t("subwave", lambda: (counts.zero_(), ops.native().cbow_eval_counts_(
-        s, g_t, o_t, l_t, p_split, counts, dO=dO, inv_b=inv_b)))
+    def run_counts():
+        counts.zero_()
+        ops.native().cbow_eval_counts_(s, g_t, o_t, l_t, p_split, counts,
+                                       dO=dO, inv_b=inv_b)
+
+    os.environ.pop("G2VEC_EVAL_LDS", None)      # default: LDS variant off
+    t("subwave", run_counts)
+    base_counts = counts.tolist()
+    base_dO = dO.clone()
     t("scan", lambda: (counts.zero_(), ops.native().cbow_eval_scan_(
         s, g_t, pathid, o_t, l_t, p_split, cap, piece, counts,
         dO=dO, inv_b=inv_b)))
+    for grid in (512, 768, 1024, 1536, 2048):
+        os.environ["G2VEC_EVAL_LDS"] = str(256 * 1024)
+        os.environ["G2VEC_EVAL_LDS_GRID"] = str(grid)
+        dO.zero_()
+        t(f"subwave+lds g{grid}", run_counts)
+        assert counts.tolist() == base_counts, (counts.tolist(), base_counts)
+        assert torch.equal(dO, base_dO), "LDS variant must be bitwise-equal"
+    os.environ.pop("G2VEC_EVAL_LDS", None)
 
 
 if __name__ == "__main__":
