@@ -240,6 +240,30 @@ def test_cbow_eval_counts_matches_oracle():
     assert float(counts[1]) == pytest.approx(float(corr[300:].sum()))
 
 
+def test_cbow_eval_counts_lds_variant_bitwise(monkeypatch):
+    """The opt-in LDS-staged eval (cbow_eval_counts_lds_kernel,
+    G2VEC_EVAL_LDS=<bytes>) must be bitwise-equal to the default subwave
+    kernel: same gathers, same subwave-16 accumulation order, s staged in
+    LDS. Off by default (measured in-chain neutral, profiles/README.md)."""
+    genes, offs, labels = _pathset_tensors(seed=9)
+    s = torch.randn(300)
+    dO_a = torch.zeros(300, device=DEV)
+    dO_b = torch.zeros(300, device=DEV)
+    counts_a = torch.zeros(2, device=DEV)
+    counts_b = torch.zeros(2, device=DEV)
+    args = (s.to(DEV), genes.to(DEV), offs.to(DEV), labels.to(DEV), 300)
+    monkeypatch.delenv("G2VEC_EVAL_LDS", raising=False)
+    ops.cbow_eval_counts_(*args, counts_a, dO=dO_a, inv_b=1.0 / 300)
+    monkeypatch.setenv("G2VEC_EVAL_LDS", str(64 * 1024))
+    for grid in ("512", "1024"):
+        monkeypatch.setenv("G2VEC_EVAL_LDS_GRID", grid)
+        counts_b.zero_()
+        dO_b.zero_()
+        ops.cbow_eval_counts_(*args, counts_b, dO=dO_b, inv_b=1.0 / 300)
+        assert torch.equal(counts_a, counts_b)
+        assert torch.equal(dO_a, dO_b)
+
+
 @pytest.mark.parametrize("G,h", [(1000, 64), (777, 128), (513, 256), (300, 512)])
 def test_gemv_kernels_match_torch(G, h):
     torch.manual_seed(3)
