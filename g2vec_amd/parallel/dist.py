@@ -149,7 +149,12 @@ def init_dist(device_hint: str = "auto") -> DistContext:
     else:
         device = torch.device("cpu")
     if world > 1 and not dist.is_initialized():
-        backend = "nccl" if use_cuda else "gloo"
+        # G2VEC_DIST_BACKEND: diagnostic override. gloo+CUDA lets a
+        # 1-GPU box rehearse the multi-process DP path (N ranks sharing
+        # one device, real kernels, CPU-staged collectives) where 2-rank
+        # RCCL on one GPU is refused ("Duplicate GPU detected").
+        backend = (os.environ.get("G2VEC_DIST_BACKEND")
+                   or ("nccl" if use_cuda else "gloo"))
         dist.init_process_group(backend=backend, rank=rank, world_size=world,
                                 timeout=datetime.timedelta(seconds=300))
     return DistContext(rank, world, device, world > 1)
