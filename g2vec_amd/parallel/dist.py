@@ -91,20 +91,47 @@ class DistContext:
             return True
         if self._graph_ok is None:
             if (os.environ.get("G2VEC_DIST_GRAPH", "1") == "0"
-                    or self.device.type != "cuda"):
+                    or self.device.type != "cuda"
+                    or dist.get_backend() != "nccl"):
+                # only RCCL kernels are capturable; probing a CPU-staged
+                # backend (gloo) would only exercise the failure path
                 self._graph_ok = False
                 return False
             t = torch.ones(2, dtype=torch.float32, device=self.device)
             dist.all_reduce(t)          # connect the communicator first:
             torch.cuda.synchronize()    # capture can't establish channels
-            g = None
-            try:
-                g = torch.cuda.CUDAGraph()
-                # thread_local: RCCL's watchdog/event threads must not
-                # invalidate the capture
-                with torch.cuda.graph(g, capture_error_mode="thread_local"):
-                    dist.all_reduce(t)
-            except Exception:  # noqa: BLE001
+            # Explicit begin/end on our OWN side stream, end in a
+            # finally: if the captured collective raises,
+            # torch.cuda.graph's __exit__ can abort before restoring the
+            # ambient stream, leaving the thread's current stream
+            # capturing — every later op then fails with
+            # "operation not permitted when stream is capturing"
+            # (observed with a non-capturable backend; the same poisoned
+            # state would follow any RCCL capture failure). "relaxed"
+            # mode so an aborted probe capture cannot invalidate
+            # unrelated work; the replay value check below is the
+            # correctness gate.
+            g = torch.cuda.CUDAGraph()
+            captured = True
+            probe_stream = torch.cuda.Stream()
+            probe_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(probe_stream):
+                try:
+                    g.capture_begin(capture_error_mode="relaxed")
+                except Exception:  # noqa: BLE001
+                    captured = False
+                else:
+                    try:
+                        dist.all_reduce(t)
+                    except Exception:  # noqa: BLE001
+                        captured = False
+                    finally:
+                        try:
+                            g.capture_end()
+                        except Exception:  # noqa: BLE001
+                            captured = False
+            torch.cuda.current_stream().wait_stream(probe_stream)
+            if not captured:
                 g = None
             # agree BEFORE replaying: if capture failed on any rank, a
             # replay elsewhere would launch a collective that rank never
