@@ -100,7 +100,8 @@ def make_real_dataset(seed: int = 0, n_common: int = N_COMMON_PUBLISHED,
                       n_modules: int = 32, lp_iters: int = 4,
                       shared_frac: float = 0.1, off_frac: float = 0.45,
                       min_module: int = 30, dead_frac: float = 0.0,
-                      ref_dir: Optional[str] = None) -> Dict:
+                      ref_dir: Optional[str] = None,
+                      sample_seed: Optional[int] = None) -> Dict:
     """Real network + real clinical + synthesized expression.
 
     Difficulty note: with the default knobs (size-ordered community ids,
@@ -157,8 +158,14 @@ def make_real_dataset(seed: int = 0, n_common: int = N_COMMON_PUBLISHED,
         dead = rng.random(n_common) < dead_frac
         module = np.where(dead, -1, module)
     expr_genes = [genes[i] for i in chosen]
-    expr = synth.synth_expression(expr_genes, raw["labels"], module, seed,
-                                  shared_frac=shared_frac, off_frac=off_frac)
+    # sample_seed (default = seed) draws only the expression sampling:
+    # DP weak scaling passes a per-rank sample_seed over a shared seed so
+    # every rank sees the SAME module ground truth (labels are the real
+    # clinical file, fixed) — see bench.build_dataset's cohort_seed note
+    expr = synth.synth_expression(
+        expr_genes, raw["labels"], module,
+        seed if sample_seed is None else sample_seed,
+        shared_frac=shared_frac, off_frac=off_frac)
     return {"expr": expr, "expr_genes": expr_genes,
             "samples": raw["samples"], "labels": raw["labels"],
             "net_genes": genes, "edge_idx": e, "module": module,
