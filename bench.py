@@ -13,12 +13,26 @@ TF1-Adam) plus the post-update train-ACC and val-ACC evaluations — exactly
 the per-epoch work of the reference (G2Vec.py:262-267), which its ~2.2
 s/epoch baseline also includes.
 
-Scaling is WEAK: each rank generates and trains its own ex_*-shaped path
-shard (global batch grows with N). Per epoch exactly ONE RCCL collective
-runs in the timed fixed-epoch region: the rank-1 backward's c vector
-(G floats; dW_ho is recomputed per rank from the reduced c, and the
-accuracy counts accumulate in a device-side history all-reduced once
-after the loop).
+Scaling is WEAK over the walk budget of ONE shared study: every rank
+builds the same dataset, the job runs num_repetition = reps x N of the
+reference algorithm (exactly "the published run with more repetitions"),
+each rank walks its source shard of all reps (C5), the walk shards are
+all-gathered for GLOBAL dedup + common-path removal, and training is
+DP-sharded over the global path set — bitwise the single-process
+trajectory at the same global batch (tests/test_dist_gloo.py). Per-GPU
+walk and training work stay ~constant as N grows; convergence IMPROVES
+with N (more repetitions = denser path coverage). Per epoch exactly ONE
+RCCL collective runs in the timed fixed-epoch region: the rank-1
+backward's c vector (G floats; dW_ho is recomputed per rank from the
+reduced c, and the accuracy counts accumulate in a device-side history
+all-reduced once after the loop).
+
+(An earlier design gave each rank an INDEPENDENT dataset; measured on a
+2-rank rehearsal, DP training over conflicting per-rank ground truths
+caps global val-ACC at ~0.85 — near-identical gene-set paths from
+different cohorts carry conflicting labels, which the reference's
+common-path removal deletes within one study but nothing can delete
+across studies. See build_dataset's cohort_seed note.)
 
 Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
 Launched multi-GPU by the driver via torch.distributed.run (one rank/GPU).
@@ -37,7 +51,7 @@ import torch
 
 from g2vec_amd.config import G2VecConfig
 from g2vec_amd.models.cbow import CbowTrainer
-from g2vec_amd.parallel.dist import init_dist, single
+from g2vec_amd.parallel.dist import init_dist
 from g2vec_amd.pipeline import generate_paths
 from g2vec_amd.utils import synth
 
@@ -52,7 +66,8 @@ def build_dataset(seed: int, n_genes: int = 7523, n_edges: int = 298799,
                   n_extra: int = 2381, n_modules: int = 16,
                   shared_frac: float = 0.2, off_frac: float = 0.55,
                   cohort_seed: int | None = None):
-    """In-memory ex_*-shaped dataset (one per rank; weak scaling).
+    """In-memory ex_*-shaped dataset (identical on every rank: the bench
+    scales the walk budget of ONE shared study, not the dataset count).
 
     shared_frac/off_frac set the convergence difficulty (calibrated with
     tools/calibrate_difficulty.py so the seeded val-ACC trajectory climbs
@@ -61,12 +76,13 @@ def build_dataset(seed: int, n_genes: int = 7523, n_edges: int = 298799,
 
     seed fixes the GROUND TRUTH (module assignment + network topology);
     cohort_seed (default = seed) draws the patient cohort (clinical
-    labels + expression sampling). Weak scaling passes a per-rank
-    cohort_seed over a shared seed: every rank's shard then carries the
-    SAME gene->class signal and DP training converges like N=1 with more
-    data — per-rank independent structure seeds gave each rank a
-    conflicting relabeling of the same gene ids, capping global val-ACC
-    below the 0.88 target (measured: DP=2 best 0.8757 in 120 epochs)."""
+    labels + expression sampling) — a diagnostics knob. DP training over
+    DIFFERENT datasets was measured and rejected for the bench: with
+    per-rank structure seeds global val-ACC caps at 0.8757, and even
+    with a shared structure but per-rank cohorts at 0.8524 (2-rank
+    rehearsal, 120 epochs) — near-identical gene-set paths from
+    different cohorts carry conflicting labels that common-path removal
+    can only delete within one study."""
     if cohort_seed is None:
         cohort_seed = seed
     rng = np.random.default_rng(seed)
@@ -138,18 +154,19 @@ def main() -> int:
     rank = ctx.rank
     log(f"[bench] rank {rank}/{world} device {device}")
 
+    # ONE study per job: the global walk budget is reps x world of the
+    # SAME dataset (weak scaling — each rank walks its source shard of
+    # all repetitions, so per-rank walk work is ~constant in N)
     cfg = G2VecConfig(hidden=args.hidden, len_path=args.len_path,
-                      num_repetition=args.reps, epochs=500, seed=args.seed,
+                      num_repetition=args.reps * world, epochs=500,
+                      seed=args.seed,
                       device=str(device.type), trainer_path=args.trainer_path,
                       dtype=args.dtype, use_hipgraph=not args.no_hipgraph)
 
     # ---- dataset + graphs + walks (input pipeline; measured, not the metric)
     if args.real_data:
         from g2vec_amd.utils import refdata
-        # same structure/cohort split as build_dataset: modules over the
-        # real topology from the base seed, per-rank expression sampling
-        ds = refdata.make_real_dataset(seed=args.seed,
-                                       sample_seed=args.seed + 1000 * rank)
+        ds = refdata.make_real_dataset(seed=args.seed)
         g2i = {g: i for i, g in enumerate(ds["net_genes"])}
         keep = np.array([g2i[g] for g in ds["expr_genes"]])
         idx_of = np.full(len(ds["net_genes"]), -1, np.int64)
@@ -164,14 +181,16 @@ def main() -> int:
         expr, labels, edge_idx, n_genes = build_dataset(
             args.seed, args.n_genes, args.n_edges, args.n_extra,
             args.n_modules, shared_frac=args.shared_frac,
-            off_frac=args.off_frac, cohort_seed=args.seed + 1000 * rank)
+            off_frac=args.off_frac)
     expr_t = torch.from_numpy(expr).to(device)
     labels_t = torch.from_numpy(labels).to(device)
     edge_t = torch.from_numpy(edge_idx).to(device)
 
-    # weak scaling: each rank walks its OWN dataset with full source
-    # coverage — a single-process context (no C5 sharding/gather here)
-    sctx = single(device)
+    # C5 sharded generation: rank r walks sources [lo, hi) of the shared
+    # study at the global repetition count; _gather_walks all-gathers the
+    # shards so every rank holds the GLOBAL walks before dedup/common
+    # removal (world 1: identical to a single-process context)
+    sctx = ctx
     if on_gpu:
         # one-time hipModule/dispatcher loads + allocator arenas for the
         # step-3 op set (~1.2 s on a fresh process,
@@ -209,7 +228,7 @@ def main() -> int:
         cfg_k = dataclasses.replace(cfg, seed=args.seed + 7919 * k)
         trainer = CbowTrainer(cfg_k, n_genes, device, ctx,
                               log=(lambda *a, **k2: None))
-        st = trainer.setup(ps, pre_sharded=True)
+        st = trainer.setup(ps, pre_sharded=False)
         wall_k = None
         acc_k = 0.0
         ep_k = None
@@ -242,7 +261,7 @@ def main() -> int:
     # ---- timed throughput region: fresh state, W warmup + K timed epochs
     trainer = CbowTrainer(cfg, n_genes, device, ctx,
                           log=(lambda *a, **k2: None))
-    st = trainer.setup(ps, pre_sharded=True)
+    st = trainer.setup(ps, pre_sharded=False)
     n_tr_global = trainer.n_tr_global
     pipelined = (on_gpu and args.trainer_path == "fast"
                  and not args.no_pipeline)
@@ -298,6 +317,7 @@ def main() -> int:
                 "parallelism": f"dp{world}",
                 "hidden": args.hidden,
                 "num_repetition": args.reps,
+                "num_repetition_global": args.reps * world,
                 "trainer_path": args.trainer_path,
                 "hipgraph_active": bool(
                     getattr(st, "graph", None) is not None
