@@ -65,3 +65,19 @@ def test_bench_torchrun_dp2_contract():
     d = json.loads(lines[0])
     assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+    # C5 weak scaling: one shared study at num_repetition = reps x world;
+    # the global batch must come from the GLOBAL dedup of both ranks'
+    # source-sharded walks (reps x world > reps of a single rank's set)
+    assert d["config"]["num_repetition_global"] == 4
+    single = subprocess.run(
+        [sys.executable, "bench.py",
+         "--n-genes", "500", "--n-edges", "6000", "--n-extra", "50",
+         "--n-modules", "6", "--reps", "2", "--len-path", "12",
+         "--steps", "2", "--warmup", "1", "--acc-target-epochs", "2"],
+        capture_output=True, text=True, timeout=400)
+    assert single.returncode == 0, single.stderr[-2000:]
+    d1 = json.loads([l for l in single.stdout.strip().splitlines()
+                     if l.strip().startswith("{")][0])
+    # same study at 2x the walk budget: strictly more unique paths, and
+    # roughly linear growth (dedup saturation would show up here)
+    assert d["config"]["global_batch"] > 1.5 * d1["config"]["global_batch"]
