@@ -51,6 +51,23 @@ def test_real_reference_counts():
     assert set(ds["expr_genes"]) <= set(ds["net_genes"])
 
 
+def test_sample_seed_varies_cohort_not_structure():
+    """make_real_dataset(sample_seed=...) must redraw ONLY the expression
+    sampling: module assignment, gene choice and edges stay fixed by
+    `seed` (the DP weak-scaling seam — bench shares one study across
+    ranks; cohort redraws over a shared structure were measured to break
+    DP convergence, profiles/dp2_rehearsal.md)."""
+    a = refdata.make_real_dataset(seed=3)
+    b = refdata.make_real_dataset(seed=3, sample_seed=3)      # default alias
+    c = refdata.make_real_dataset(seed=3, sample_seed=99)
+    assert np.array_equal(a["expr"], b["expr"])
+    assert not np.array_equal(a["expr"], c["expr"])
+    assert np.array_equal(a["module"], c["module"])
+    assert a["expr_genes"] == c["expr_genes"]
+    assert np.array_equal(a["edge_idx"], c["edge_idx"])
+    assert np.array_equal(a["labels"], c["labels"])           # real clinical
+
+
 @pytest.mark.skipif(not HAVE_REF, reason="/root/reference not mounted")
 def test_cache_matches_reference_files():
     """The committed npz cache is byte-equivalent to parsing the real
