@@ -132,7 +132,8 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
   if (G * 4 <= lds_cap && G * 4 <= 158 * 1024) {   // 160 KB LDS hard limit
     const char* ge = getenv("G2VEC_EVAL_LDS_GRID");
     int grid = grid_for(P, 16);
-    const int gcap = ge ? atoi(ge) : 1024;
+    int gcap = ge ? atoi(ge) : 1024;
+    if (gcap < 1) gcap = 1;         // malformed env must not 0-block-launch
     if (grid > gcap) grid = gcap;
     auto partials = torch::empty({grid, 2},
         torch::TensorOptions().dtype(at::kFloat).device(s.device()));

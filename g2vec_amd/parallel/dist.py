@@ -130,8 +130,13 @@ class DistContext:
                             g.capture_end()
                         except Exception:  # noqa: BLE001
                             captured = False
-            torch.cuda.current_stream().wait_stream(probe_stream)
-            if not captured:
+            if captured:
+                # ordering hygiene only — capture does not execute work,
+                # so on failure nothing on probe_stream needs waiting on
+                # (and waiting on a stream whose capture_end failed could
+                # itself raise)
+                torch.cuda.current_stream().wait_stream(probe_stream)
+            else:
                 g = None
             # agree BEFORE replaying: if capture failed on any rank, a
             # replay elsewhere would launch a collective that rank never
