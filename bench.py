@@ -268,7 +268,9 @@ def main() -> int:
     if pipelined:
         trainer.run_epochs_pipelined(st, args.warmup, early_stop=False)
         if trainer.kblock_eligible(st, args.steps, False) and st.epoch_idx:
-            trainer._ensure_kgraph(st)      # one-time capture stays untimed
+            # one-time capture stays untimed; block size tiles the timed
+            # run so no epoch falls to the slower eager tail
+            trainer._ensure_kgraph(st, k=trainer.pick_kblock(args.steps))
     else:
         for _ in range(args.warmup):
             trainer.run_epoch(st)
