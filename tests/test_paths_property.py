@@ -1,7 +1,11 @@
 """Property-based checks of path-set integration (hypothesis)."""
+import os
+
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st
+
+_SOAK = max(int(os.environ.get("G2VEC_SOAK", "1")), 1)  # soak runs scale the example budget
 
 from g2vec_amd.ops.cpu_ref import path_hash
 from g2vec_amd.paths import integrate_pathsets
@@ -24,7 +28,7 @@ def _ws(paths, len_path=6):
     return WalkSet(nodes, lengths, hashes)
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60 * _SOAK, deadline=None)
 @given(good=paths_strategy, poor=paths_strategy)
 def test_integrate_properties(good, poor):
     ps, freq, n_in = integrate_pathsets(_ws(good), _ws(poor), 20)

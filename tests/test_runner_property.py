@@ -1,11 +1,15 @@
 """Property-based checks of the epoch runners (hypothesis): for random
 path sets, epoch budgets, granularities and seeds, every runner variant
 must reproduce the synchronous reference loop exactly."""
+import os
+
 import numpy as np
 import pytest
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st
+
+_SOAK = max(int(os.environ.get("G2VEC_SOAK", "1")), 1)  # soak runs scale the example budget
 
 from g2vec_amd.config import G2VecConfig
 from g2vec_amd.models.cbow import CbowTrainer
@@ -31,7 +35,7 @@ def _train_sync(cfg, ps):
     return tr, tr.train(ps)
 
 
-@settings(max_examples=12, deadline=None)
+@settings(max_examples=12 * _SOAK, deadline=None)
 @given(seed=st.integers(0, 50), epochs=st.integers(2, 25),
        k=st.integers(2, 9), data_seed=st.integers(0, 20))
 def test_kgranular_property(seed, epochs, k, data_seed):
@@ -48,7 +52,7 @@ def test_kgranular_property(seed, epochs, k, data_seed):
     assert torch.equal(W, ref.W_ih)
 
 
-@settings(max_examples=10, deadline=None)
+@settings(max_examples=10 * _SOAK, deadline=None)
 @given(seed=st.integers(0, 50), epochs=st.integers(3, 20),
        cut=st.floats(0.2, 0.8), data_seed=st.integers(0, 20))
 def test_checkpoint_resume_property(tmp_path_factory, seed, epochs, cut,
@@ -67,7 +71,7 @@ def test_checkpoint_resume_property(tmp_path_factory, seed, epochs, cut,
     assert torch.equal(resumed.W_ih, full.W_ih)
 
 
-@settings(max_examples=10, deadline=None)
+@settings(max_examples=10 * _SOAK, deadline=None)
 @given(seed=st.integers(0, 50), epochs=st.integers(2, 18),
        data_seed=st.integers(0, 20))
 def test_pipelined_property(seed, epochs, data_seed):
