@@ -354,3 +354,21 @@ def test_activation_requires_general_path():
         G2VecConfig(activation="relu").validate()
     with _pt.raises(ValueError, match="activation"):
         G2VecConfig(activation="gelu", trainer_path="general").validate()
+
+
+def test_pick_kblock_tiles_run_length():
+    """pick_kblock must return a block size whose replays tile the run
+    (no eager-tail epochs): n itself when small, else the largest
+    divisor <= 64, else the default KBLOCK for awkward lengths."""
+    cfg = G2VecConfig(hidden=8, epochs=5, seed=0, device="cpu")
+    tr = CbowTrainer(cfg, 10, torch.device("cpu"), log=lambda *a, **k: None)
+    assert tr.pick_kblock(30) == 30
+    assert tr.pick_kblock(64) == 64
+    assert tr.pick_kblock(1) == 1
+    assert tr.pick_kblock(0) == 1
+    assert tr.pick_kblock(100) == 50
+    assert tr.pick_kblock(200) == 50
+    assert tr.pick_kblock(500) == 50
+    for n in (30, 64, 100, 200, 500):
+        assert n % tr.pick_kblock(n) == 0
+    assert tr.pick_kblock(67) == tr.KBLOCK      # prime > 64: short tail
