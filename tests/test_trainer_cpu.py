@@ -360,7 +360,7 @@ def test_pick_kblock_tiles_run_length():
     """pick_kblock must return a block size whose replays tile the run
     (no eager-tail epochs): n itself when small, else the largest
     divisor <= 64, else the default KBLOCK for awkward lengths."""
-    cfg = G2VecConfig(hidden=8, epochs=5, seed=0, device="cpu")
+    cfg = G2VecConfig(hidden=64, epochs=5, seed=0, device="cpu")
     tr = CbowTrainer(cfg, 10, torch.device("cpu"), log=lambda *a, **k: None)
     assert tr.pick_kblock(30) == 30
     assert tr.pick_kblock(64) == 64
