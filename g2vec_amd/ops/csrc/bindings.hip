@@ -152,7 +152,10 @@ void cbow_eval_counts_(torch::Tensor s, torch::Tensor genes, torch::Tensor offs,
     return;
   }
   int grid = grid_for(P, 16);       // 16 paths per block (16-lane sub-waves)
-  if (grid > 2048) grid = 2048;      // 8192 waves fill the chip
+  const char* gge = getenv("G2VEC_EVAL_GRID");   // sweep knob; default
+  int gcap2 = gge ? atoi(gge) : 2048;            // 8192 waves fill the chip
+  if (gcap2 < 1) gcap2 = 1;
+  if (grid > gcap2) grid = gcap2;
   auto partials = torch::empty({grid, 2},
       torch::TensorOptions().dtype(at::kFloat).device(s.device()));
   hipLaunchKernelGGL(cbow_eval_counts_kernel, dim3(grid),
