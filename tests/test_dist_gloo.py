@@ -425,3 +425,44 @@ def test_dp2_general_path_matches_single_process():
         W_dp, hist_dp = got[act]
         assert hist_dp == pytest.approx(res.acc_val_history, abs=1e-6), act
         assert np.allclose(W_dp, res.W_ih.numpy(), atol=1e-5), act
+
+
+def _unseeded_worker(rank, world, port, q):
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        ctx = DistContext(rank, world, torch.device("cpu"), True)
+        ps = _pathset()
+        cfg = G2VecConfig(hidden=64, epochs=4, early_stop=False, seed=None,
+                          device="cpu", dtype="fp32")
+        tr = CbowTrainer(cfg, ps.n_genes, torch.device("cpu"), ctx,
+                         log=lambda *a, **k: None)
+        res = tr.train(ps)
+        q.put((rank, res.acc_val_history, res.W_ih.numpy()))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_unseeded_ranks_agree():
+    """seed=None at world>1: ranks must broadcast-agree on the global
+    shuffle seed and the weight init (rank-0 broadcast), or the shards
+    would overlap/miss paths and the trajectories diverge. Asserts both
+    ranks return identical histories and weights (the reduced trajectory
+    is global, so agreement proves the collectives lined up)."""
+    port = _free_port()
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_unseeded_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(2):
+        rank, hist, W = q.get(timeout=240)
+        got[rank] = (hist, W)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got[0][0] == got[1][0]
+    assert np.array_equal(got[0][1], got[1][1])
