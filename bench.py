@@ -102,7 +102,41 @@ def build_dataset(seed: int, n_genes: int = 7523, n_edges: int = 298799,
     return expr, np.asarray(labels), edge_idx.astype(np.int32), n_genes
 
 
+def _launch_health_probe() -> None:
+    """~8% of fresh processes on this pool come up with a degraded HIP
+    submission path: every launch/copy costs ~0.3 ms instead of ~3 us,
+    so the 0.064 ms epoch measures ~2.4 ms (observed twice in ~25 runs,
+    profiles/README.md; the same box is normal in the neighboring
+    processes). Detect it BEFORE any measurement by timing 200 trivial
+    launches; if degraded, re-exec once — the replacement process
+    re-initializes HIP cleanly. Runs before init_dist so a re-exec'd
+    torchrun rank simply rejoins the rendezvous."""
+    import os
+    if os.environ.get("G2VEC_HEALTH_REEXEC") == "1":
+        return
+    if not torch.cuda.is_available():
+        return
+    t = torch.zeros(1, device="cuda")
+    for _ in range(5):
+        t.fill_(1.0)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(200):
+        t.fill_(1.0)
+    torch.cuda.synchronize()
+    per_launch = (time.perf_counter() - t0) / 200
+    if per_launch > 100e-6:
+        log(f"[bench] WARNING: degraded HIP launch path "
+            f"({per_launch * 1e6:.0f} us/launch vs ~3 normal); "
+            f"re-executing once for a clean process")
+        os.environ["G2VEC_HEALTH_REEXEC"] = "1"
+        sys.stderr.flush()
+        sys.stdout.flush()
+        os.execv(sys.executable, [sys.executable] + sys.argv)
+
+
 def main() -> int:
+    _launch_health_probe()
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=30)
