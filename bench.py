@@ -116,7 +116,11 @@ def _launch_health_probe() -> None:
         return
     if not torch.cuda.is_available():
         return
-    t = torch.zeros(1, device="cuda")
+    # probe THIS rank's GPU (LOCAL_RANK not yet consumed by init_dist) —
+    # probing cuda:0 from every rank would pin 8 stray contexts there
+    dev = int(os.environ.get("LOCAL_RANK", "0")) % torch.cuda.device_count()
+    torch.cuda.set_device(dev)
+    t = torch.zeros(1, device=f"cuda:{dev}")
     for _ in range(5):
         t.fill_(1.0)
     torch.cuda.synchronize()
