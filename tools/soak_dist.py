@@ -2,9 +2,10 @@
 dense walker over 20 random graph shapes (TV distance of length
 histograms + coverage correlation)."""
 import sys
-_root = __file__.rsplit('/', 2)[0]
+import os
+_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, _root)
-sys.path.insert(0, _root + '/tests')
+sys.path.insert(0, os.path.join(_root, 'tests'))
 import numpy as np, torch
 from g2vec_amd import ops
 from g2vec_amd.graph import build_group_graph
@@ -20,7 +21,8 @@ def csr_to_dense(g):
     return adj
 
 rng = np.random.default_rng(5)
-for rnd in range(20):
+N_ROUNDS = int(os.environ.get('SOAK_ROUNDS', '20'))
+for rnd in range(N_ROUNDS):
     G = int(rng.integers(60, 220))
     lp = int(rng.integers(8, 30))
     ds = synth.synth_dataset(G, int(G * rng.integers(6, 15)), 60,
@@ -60,4 +62,4 @@ for rnd in range(20):
     assert r > 0.97, (rnd, r)
     assert dm < 0.6, (rnd, dm)
     print(f"round {rnd}: G={G} lp={lp} tv={tv:.4f} cov_r={r:.4f} dmean={dm:.3f} OK", flush=True)
-print("DIST SOAK PASS: 20 rounds")
+print(f"DIST SOAK PASS: {N_ROUNDS} rounds")

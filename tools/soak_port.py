@@ -1,9 +1,10 @@
 """Ad-hoc soak: framework integrate/freq vs literal port over many random
 walk multisets and graph shapes (50 rounds)."""
 import sys
-_root = __file__.rsplit('/', 2)[0]
+import os
+_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, _root)
-sys.path.insert(0, _root + '/tests')
+sys.path.insert(0, os.path.join(_root, 'tests'))
 import numpy as np, torch
 from g2vec_amd import ops
 from g2vec_amd.graph import build_group_graph
@@ -20,7 +21,8 @@ def walkset_to_paths(w):
     return out
 
 rng = np.random.default_rng(2)
-for rnd in range(50):
+N_ROUNDS = int(os.environ.get('SOAK_ROUNDS', '50'))
+for rnd in range(N_ROUNDS):
     G = int(rng.integers(40, 300))
     ds = synth.synth_dataset(G, int(G * rng.integers(5, 20)), 60,
                              n_modules=int(rng.integers(3, 9)),
@@ -56,4 +58,4 @@ for rnd in range(50):
         assert f[i] == port_freq.get(nm, 2), (rnd, nm)
     assert n_gip == len(port_freq), rnd
     print(f"round {rnd}: G={G} paths={ps.n_paths} OK", flush=True)
-print("SOAK PASS: 50 rounds")
+print(f"SOAK PASS: {N_ROUNDS} rounds")
